@@ -1,0 +1,191 @@
+"""Evaluation driver: batched predictor, NMS, txt emission, demo.
+
+Parity with /root/reference/evaluate.py:15-290: ``single_device_evaluate``
+runs the test split, rescales boxes to the original image size from the voc
+dict, pickles the results and writes the per-image ``txt/`` files
+(``"%d %f %d %d %d %d"``) that the external VOC-mAP tool consumed — plus, in
+this rebuild, the mAP itself is computed in-repo (``metrics.py``; the
+reference delegated it to an out-of-process submodule, README.md:42-44).
+
+MI355X re-design: ``Prediction`` decodes ALL batch items and ALL stacks in
+one fused batched decode (peak-mask + top-k + gather — the HIP kernel on
+GPU) instead of a per-item, per-stack python loop around torch primitives.
+NMS then runs per item over the concatenated stacks, same contract as the
+reference (class-agnostic, evaluate.py:174).
+"""
+
+import os
+import time
+from collections import defaultdict
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..utils import AverageMeter, save_pickle
+from ..data import load_dataset
+from .trainer import load_network
+
+try:
+    from tqdm import tqdm
+except ImportError:  # pragma: no cover
+    def tqdm(x):
+        return x
+
+
+class Prediction(torch.nn.Module):
+    """network forward -> sigmoid -> batched decode -> per-item NMS."""
+
+    def __init__(self, network, topk, scale_factor, conf_th, nms, nms_th,
+                 normalized_coord=False, pool_size=3):
+        super().__init__()
+        self.network = network
+        self.topk = topk
+        self.scale_factor = scale_factor
+        self.conf_th = conf_th
+        self.nms = nms
+        self.nms_th = nms_th
+        self.normalized_coord = normalized_coord
+        self.pool_size = pool_size
+
+    @torch.no_grad()
+    def forward(self, x):
+        batch_output = self.network(x)  # (B, S, num_cls+4, h, w)
+        b, s, c, h, w = batch_output.shape
+        num_cls = c - 4
+
+        flat = batch_output.reshape(b * s, c, h, w).float()
+        heatmap, offset, wh = flat.split([num_cls, 2, 2], dim=1)
+        heatmap = torch.sigmoid(heatmap)
+        if self.normalized_coord:
+            offset = torch.sigmoid(offset)
+            wh = torch.sigmoid(wh)
+
+        boxes, clss, scores = ops.batched_decode(
+            heatmap, offset, wh, self.scale_factor, self.topk,
+            self.pool_size, self.normalized_coord)
+        # merge stacks: (B, S*topk, ...) — deep-supervised stacks all vote,
+        # NMS dedups (reference evaluate.py:133-158)
+        boxes = boxes.reshape(b, s * self.topk, 4)
+        clss = clss.reshape(b, s * self.topk)
+        scores = scores.reshape(b, s * self.topk)
+
+        box_lst, cls_lst, score_lst = [], [], []
+        for i in range(b):
+            keep = scores[i] >= self.conf_th
+            bi, ci, si = boxes[i][keep], clss[i][keep], scores[i][keep]
+            bi2, ci2, si2 = self.nonmaximum_supression(bi, ci, si)
+            box_lst.append(bi2)
+            cls_lst.append(ci2)
+            score_lst.append(si2)
+        return box_lst, cls_lst, score_lst
+
+    def nonmaximum_supression(self, boxes, clss, scores):
+        if self.nms == 'nms':
+            keep = ops.nms(boxes, scores, self.nms_th)
+        elif self.nms == 'soft-nms':
+            keep, rescored = ops.soft_nms(boxes, scores,
+                                          score_th=self.conf_th)
+            return boxes[keep], clss[keep], rescored
+        else:
+            raise NotImplementedError(
+                'Not expected nms algorithm: %s' % self.nms)
+        return boxes[keep], clss[keep], scores[keep]
+
+
+def resize_box_to_original_scale(boxes, original_size, transformed_size):
+    """Scale (N,4) xyxy from the network input size back to the source image."""
+    ow, oh = original_size
+    tw, th = transformed_size
+    boxes = np.asarray(boxes, dtype=np.float64).reshape(-1, 4)
+    out = boxes.copy()
+    out[:, [0, 2]] *= ow / tw
+    out[:, [1, 3]] *= oh / th
+    return out
+
+
+def evaluate_step(dataloader, predictor, device, args):
+    """Run the predictor over the loader -> {filename: (N,6) array}."""
+    time_logger = defaultdict(AverageMeter)
+    prediction_results = {}
+    predictor.eval()
+    imsize = args.imsize or args.multiscale[1]
+
+    tictoc = time.time()
+    for image, gt_heatmap, gt_offset, gt_size, gt_mask, gt_dict in \
+            tqdm(dataloader):
+        time_logger['data'].update(time.time() - tictoc)
+
+        tictoc = time.time()
+        box_lst, cls_lst, score_lst = predictor(image.to(device))
+        time_logger['forward'].update(time.time() - tictoc)
+
+        for b in range(image.shape[0]):
+            gt_info = gt_dict[b]
+            size = gt_info['annotation']['size']
+            origin_size = int(size['width']), int(size['height'])
+            boxes = resize_box_to_original_scale(
+                box_lst[b].detach().cpu().numpy(), origin_size,
+                (imsize, imsize))
+            clss = cls_lst[b].detach().cpu().numpy()[:, None]
+            scores = score_lst[b].detach().cpu().numpy()[:, None]
+            if boxes.shape[0]:
+                pred = np.hstack([clss, scores, boxes])
+            else:
+                pred = np.zeros((0, 6))
+            prediction_results[gt_info['annotation']['filename']] = pred
+        tictoc = time.time()
+
+    print('%s: Evaluation, Time(ms) [data: %6.2f, forward: %6.2f]'
+          % (time.ctime(), time_logger['data'].avg * 1e3,
+             time_logger['forward'].avg * 1e3))
+    return prediction_results
+
+
+def write_detection_txt(save_path, predictions):
+    """Per-image txt files in the mAP-tool format (evaluate.py:46-54)."""
+    os.makedirs(save_path, exist_ok=True)
+    for filename, prediction in predictions.items():
+        cls_ids, scores, boxes = (prediction[:, 0], prediction[:, 1],
+                                  prediction[:, 2:])
+        out = os.path.join(save_path, os.path.splitext(filename)[0] + '.txt')
+        with open(out, 'w') as f:
+            for i in range(cls_ids.shape[0]):
+                f.write('%d %f %d %d %d %d\n'
+                        % (cls_ids[i], scores[i], boxes[i][0], boxes[i][1],
+                           boxes[i][2], boxes[i][3]))
+
+
+def single_device_evaluate(args):
+    """Full eval pass (reference evaluate.py:15-56) + in-repo mAP."""
+    device = torch.device('cpu' if -1 in args.gpu_no else 'cuda')
+    print('%s: Use %s for evaluation'
+          % (time.ctime(),
+             'CPU' if -1 in args.gpu_no else 'GPU %d' % args.gpu_no[0]))
+
+    network, _, _, _ = load_network(args, device)
+    predictor = Prediction(
+        network=network, topk=args.topk, scale_factor=args.scale_factor,
+        conf_th=args.conf_th, nms=args.nms, nms_th=args.nms_th,
+        normalized_coord=args.normalized_coord,
+        pool_size=args.pool_size).to(device)
+
+    dataset = load_dataset(args)
+    dataloader = torch.utils.data.DataLoader(
+        dataset=dataset, batch_size=args.batch_size, shuffle=False,
+        num_workers=args.num_workers, collate_fn=dataset.collate_fn)
+
+    predictions = evaluate_step(dataloader, predictor, device, args)
+
+    save_pickle(os.path.join(args.save_path, 'prediction_results.pickle'),
+                predictions)
+    write_detection_txt(os.path.join(args.save_path, 'txt'), predictions)
+
+    # in-repo VOC mAP (reference used the external mAP submodule)
+    from .metrics import voc_map_from_dataset
+    result = voc_map_from_dataset(dataset, predictions)
+    if result is not None:
+        for cls_name, ap in sorted(result['ap'].items()):
+            print('%s: AP[%s] = %.2f%%' % (time.ctime(), cls_name, ap * 100))
+        print('%s: mAP = %.2f%%' % (time.ctime(), result['map'] * 100))
+    return predictions

@@ -1,0 +1,120 @@
+"""TorchScript export path (parity with /root/reference/export.py:8-152).
+
+``Export`` is the trace-friendly predictor variant: fixed batch=1, static
+top-k decode, and a @torch.jit.script greedy NMS (loops preserved
+symbolically, not trace-unrolled) so the traced module is self-contained —
+the C++ inference app (tools/cpp_infer) loads it with LibTorch and needs no
+python. ``export_model`` traces and saves the cpu/gpu variants
+(jit_traced_model_cpu.pth / jit_traced_model_gpu.pth, export.py:120-130).
+"""
+
+import torch
+
+from ..transform import hm2box
+
+
+@torch.jit.script
+def nms_scripted(boxes: torch.Tensor, scores: torch.Tensor,
+                 threshold: float) -> torch.Tensor:
+    """Greedy IoU suppression, scripted so the loop survives tracing."""
+    order = torch.argsort(scores, descending=True)
+    b = boxes[order]
+    s = scores[order].clone()
+    n = b.shape[0]
+    area = (b[:, 2] - b[:, 0]).clamp(min=0.) * (b[:, 3] - b[:, 1]).clamp(min=0.)
+    for i in range(n - 1):
+        if float(s[i]) == 0.:
+            continue
+        xx1 = torch.max(b[i, 0], b[i + 1:, 0])
+        yy1 = torch.max(b[i, 1], b[i + 1:, 1])
+        xx2 = torch.min(b[i, 2], b[i + 1:, 2])
+        yy2 = torch.min(b[i, 3], b[i + 1:, 3])
+        w = (xx2 - xx1).clamp(min=0.)
+        h = (yy2 - yy1).clamp(min=0.)
+        inter = w * h
+        iou = inter / (area[i] + area[i + 1:] - inter).clamp(min=1e-9)
+        s[i + 1:] = s[i + 1:] * (iou < threshold).to(s.dtype)
+    return order[s > 0]
+
+
+class Export(torch.nn.Module):
+    """Traceable single-image predictor: net -> sigmoid -> decode -> NMS.
+
+    Unlike the reference (which hardcoded split([2,2,2]) i.e. num_cls=2,
+    export.py:31) the class count is taken from the constructor.
+    Returns (boxes, classes, scores) for the first batch item.
+    """
+
+    def __init__(self, network, topk, scale_factor, conf_th, nms_th,
+                 normalized_coord=False, num_cls=2, pool_size=3):
+        super().__init__()
+        self.network = network
+        self.topk = topk
+        self.scale_factor = scale_factor
+        self.conf_th = conf_th
+        self.nms_th = nms_th
+        self.normalized_coord = normalized_coord
+        self.num_cls = num_cls
+        self.pool_size = pool_size
+
+    def forward(self, x):
+        batch_output = self.network(x)  # (1, S, num_cls+4, h, w)
+        outputs = batch_output[0]  # (S, num_cls+4, h, w)
+        stack_boxes = []
+        stack_clss = []
+        stack_scores = []
+        for output in outputs.split(1, dim=0):
+            out = output.squeeze(0).float()
+            heatmap, offset, wh = out.split([self.num_cls, 2, 2], dim=0)
+            heatmap = torch.sigmoid(heatmap)
+            if self.normalized_coord:
+                offset = torch.sigmoid(offset)
+                wh = torch.sigmoid(wh)
+            boxes, clss, scores = hm2box(
+                heatmap=heatmap, offset=offset, wh=wh,
+                scale_factor=self.scale_factor, topk=self.topk,
+                conf_th=self.conf_th, normalized=self.normalized_coord,
+                pool_size=self.pool_size)
+            stack_boxes.append(boxes)
+            stack_clss.append(clss)
+            stack_scores.append(scores)
+        boxes = torch.cat(stack_boxes, dim=0)
+        clss = torch.cat(stack_clss, dim=0)
+        scores = torch.cat(stack_scores, dim=0)
+        keep = nms_scripted(boxes, scores, self.nms_th)
+        return boxes[keep], clss[keep], scores[keep]
+
+
+def build_export_module(args, network):
+    return Export(
+        network=network, topk=args.topk, scale_factor=args.scale_factor,
+        conf_th=args.conf_th, nms_th=args.nms_th,
+        normalized_coord=args.normalized_coord, num_cls=args.num_cls,
+        pool_size=args.pool_size)
+
+
+def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
+    """Trace and save cpu (and, if available, gpu) TorchScript models."""
+    import os
+    predictor.eval()
+    paths = {}
+
+    x = torch.randn(1, 3, imsize, imsize)
+    with torch.no_grad():
+        traced_cpu = torch.jit.trace(predictor.cpu(), x)
+    p = os.path.join(save_dir, 'jit_traced_model_cpu.pth')
+    torch.jit.save(traced_cpu, p)
+    paths['cpu'] = p
+    print('Model saved at cpu:', p)
+
+    if do_gpu is None:
+        do_gpu = torch.cuda.is_available()
+    if do_gpu:
+        xg = torch.randn(1, 3, imsize, imsize, device='cuda')
+        with torch.no_grad():
+            traced_gpu = torch.jit.trace(predictor.cuda(), xg)
+        p = os.path.join(save_dir, 'jit_traced_model_gpu.pth')
+        torch.jit.save(traced_gpu, p)
+        paths['gpu'] = p
+        print('Model saved at gpu:', p)
+    return paths

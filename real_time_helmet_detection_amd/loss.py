@@ -1,0 +1,123 @@
+"""CenterNet losses: penalty-reduced focal + masked normed-L1.
+
+Math contract from /root/reference/loss.py:6-69:
+
+- FocalLoss(pred, gt, mask): ``-(mean_b sum_chw[ log(p+eps)(1-p)^a * mask ]
+  + mean_b sum_chw[ log(1-p+eps) p^a (1-gt)^b (1-mask) ]) / num_pos`` with
+  num_pos = mask.sum().clamp(1, 1e30). The (B,1,h,w) mask broadcasts over
+  the class channels.
+- NormedL1Loss(pred, gt, mask): ``mean_b sum_chw |pred*mask - gt*mask| /
+  num_pos``.
+- LossCalculator weights them (hm/offset/size) and keeps a loss history
+  ``.log`` dict that is checkpointed (reference train.py:82).
+
+MI355X design difference: the reference calls ``.item()`` on every partial
+loss every iteration (loss.py:27-30) which forces a device sync per step.
+Here the history is accumulated as detached device scalars and flushed with a
+single sync only when the log is read (``get_log``/``state_dict``), keeping
+the hot loop sync-free. On GPU the three losses are computed by one fused
+HIP kernel (ops.centernet_loss) instead of ~20 eager elementwise kernels.
+"""
+
+import torch
+import torch.nn as nn
+
+from . import ops
+
+
+def focal_loss_eager(pred, gt, mask, alpha, beta, eps=1e-7):
+    """Penalty-reduced pixelwise focal loss (see module docstring)."""
+    neg_inds = 1.0 - mask
+    neg_weights = torch.pow(1.0 - gt, beta)
+    pos_loss = torch.log(pred + eps) * torch.pow(1.0 - pred, alpha) * mask
+    neg_loss = (torch.log(1.0 - pred + eps) * torch.pow(pred, alpha)
+                * neg_weights * neg_inds)
+    pos = pos_loss.sum(dim=[1, 2, 3]).mean()
+    neg = neg_loss.sum(dim=[1, 2, 3]).mean()
+    num_pos = mask.sum().clamp(1, 1e30)
+    return -(pos + neg) / num_pos
+
+
+def normed_l1_loss_eager(pred, gt, mask):
+    loss = torch.abs(pred * mask - gt * mask)
+    loss = loss.sum(dim=[1, 2, 3]).mean()
+    num_pos = mask.sum().clamp(1, 1e30)
+    return loss / num_pos
+
+
+class FocalLoss(nn.Module):
+    def __init__(self, alpha=2.0, beta=4.0):
+        super().__init__()
+        self.alpha = alpha
+        self.beta = beta
+
+    def forward(self, pred, gt, mask, eps=1e-7):
+        return focal_loss_eager(pred, gt, mask, self.alpha, self.beta, eps)
+
+
+class NormedL1Loss(nn.Module):
+    def forward(self, pred, gt, mask):
+        return normed_l1_loss_eager(pred, gt, mask)
+
+
+class LossCalculator(nn.Module):
+    """Weighted sum of the three losses + sync-free history log."""
+
+    LOG_KEYS = ('hm', 'offset', 'size', 'total')
+
+    def __init__(self, hm_weight=1.0, offset_weight=1.0, size_weight=0.1,
+                 focal_alpha=2.0, focal_beta=4.0):
+        super().__init__()
+        self.log = {k: [] for k in self.LOG_KEYS}
+        self._pending = []  # list of detached (hm, off, size, total) tuples
+        self.hm_weight = hm_weight
+        self.offset_weight = offset_weight
+        self.size_weight = size_weight
+        self.focal_alpha = focal_alpha
+        self.focal_beta = focal_beta
+
+    def forward(self, phm, poff, psize, ghm, goff, gsize, mask):
+        hm_loss, offset_loss, size_loss = ops.centernet_losses(
+            phm, poff, psize, ghm, goff, gsize, mask,
+            self.focal_alpha, self.focal_beta)
+        total_loss = (hm_loss * self.hm_weight
+                      + offset_loss * self.offset_weight
+                      + size_loss * self.size_weight)
+        self._pending.append((hm_loss.detach(), offset_loss.detach(),
+                              size_loss.detach(), total_loss.detach()))
+        return total_loss
+
+    def flush_log(self):
+        """Materialize pending device scalars into ``.log`` (one sync)."""
+        if not self._pending:
+            return
+        stacked = torch.stack([torch.stack(t) for t in self._pending])
+        vals = stacked.cpu().tolist()
+        for row in vals:
+            for key, v in zip(self.LOG_KEYS, row):
+                self.log[key].append(v)
+        self._pending = []
+
+    def get_log(self, length=100):
+        self.flush_log()
+        parts = []
+        for key in self.LOG_KEYS:
+            hist = self.log[key]
+            n = min(length, len(hist))
+            avg = sum(hist[-n:]) / n if n else float('nan')
+            parts.append('%s: %5.2f' % (key, avg))
+        return ', '.join(parts)
+
+    # keep .log serializable in checkpoints even if callers grab it directly
+    def state_dict(self, *a, **kw):
+        self.flush_log()
+        sd = super().state_dict(*a, **kw)
+        return sd
+
+    def get_loss_log(self):
+        self.flush_log()
+        return self.log
+
+    def load_loss_log(self, log):
+        if log:
+            self.log = {k: list(v) for k, v in log.items()}

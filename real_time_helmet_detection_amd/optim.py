@@ -1,0 +1,27 @@
+"""Optimizer/scheduler factory (reference /root/reference/optim.py:3-12).
+
+Adam at ``--lr`` plus MultiStepLR(milestones, gamma). ``--optim`` selects the
+algorithm; Adam is the reference default and SGD/AdamW are accepted for
+experiments (same signature either way).
+"""
+
+import torch.optim as optim
+
+
+def get_optimizer(network, lr, lr_milestone, lr_gamma, algo='Adam'):
+    algo = (algo or 'Adam').lower()
+    params = network.parameters()
+    if algo == 'adam':
+        optimizer = optim.Adam(params, lr=lr)
+    elif algo == 'adamw':
+        optimizer = optim.AdamW(params, lr=lr)
+    elif algo == 'sgd':
+        optimizer = optim.SGD(params, lr=lr, momentum=0.9)
+    else:
+        raise ValueError(f'unknown optimizer {algo!r}')
+
+    scheduler = None
+    if lr_milestone is not None:
+        scheduler = optim.lr_scheduler.MultiStepLR(
+            optimizer=optimizer, milestones=lr_milestone, gamma=lr_gamma)
+    return optimizer, scheduler

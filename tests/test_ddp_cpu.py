@@ -1,0 +1,129 @@
+"""Multi-process CPU (gloo) tests of the bucketed gradient sync — the same
+code path the 8-GPU RCCL run uses (parallel/ddp.py)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from real_time_helmet_detection_amd.models import StackedHourglass
+
+
+def _build_net(seed=0):
+    torch.manual_seed(seed)
+    return StackedHourglass(num_stack=1, in_ch=8, out_ch=6)
+
+
+def _rank_input(rank, seed=42):
+    g = torch.Generator().manual_seed(seed + rank)
+    return torch.randn(2, 3, 64, 64, generator=g)
+
+
+def _expected_mean_grads(world=2):
+    """Single-process oracle: mean over ranks of per-rank gradients."""
+    grads = None
+    for rank in range(world):
+        net = _build_net()
+        net(_rank_input(rank)).sum().backward()
+        g = [p.grad.clone() for p in net.parameters()]
+        grads = g if grads is None else [a + b for a, b in zip(grads, g)]
+    return [g / world for g in grads]
+
+
+def _worker(rank, world, rdv_file, bucket_cap_mb, result_dir):
+    from real_time_helmet_detection_amd.parallel.ddp import \
+        BucketedDataParallel
+    dist.init_process_group('gloo', init_method=f'file://{rdv_file}',
+                            world_size=world, rank=rank)
+    try:
+        net = _build_net(seed=rank * 7)  # different init per rank on purpose
+        ddp = BucketedDataParallel(net, bucket_cap_mb=bucket_cap_mb)
+        # broadcast must have made params identical to rank 0's init
+        ddp(_rank_input(rank)).sum().backward()
+        ddp.finish_backward()
+        if rank == 0:
+            torch.save([p.grad for p in net.parameters()],
+                       os.path.join(result_dir, 'grads.pt'))
+            torch.save(net.state_dict(),
+                       os.path.join(result_dir, 'params.pt'))
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def _worker_no_sync(rank, world, rdv_file, result_dir):
+    from real_time_helmet_detection_amd.parallel.ddp import \
+        BucketedDataParallel
+    dist.init_process_group('gloo', init_method=f'file://{rdv_file}',
+                            world_size=world, rank=rank)
+    try:
+        net = _build_net()
+        ddp = BucketedDataParallel(net, bucket_cap_mb=0.25)
+        with ddp.no_sync():
+            ddp(_rank_input(rank)).sum().backward()
+        ddp.finish_backward()  # no pending comm
+        if rank == 1:
+            torch.save([p.grad for p in net.parameters()],
+                       os.path.join(result_dir, 'grads_nosync.pt'))
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize('bucket_cap_mb', [0.125, 64.0])
+def test_bucketed_allreduce_matches_oracle(tmp_path, bucket_cap_mb):
+    world = 2
+    rdv = tmp_path / f'rdv_{bucket_cap_mb}'
+    mp.spawn(_worker, nprocs=world,
+             args=(world, str(rdv), bucket_cap_mb, str(tmp_path)),
+             join=True)
+    got = torch.load(tmp_path / 'grads.pt', weights_only=False)
+    want = _expected_mean_grads(world)
+    assert len(got) == len(want)
+    for g, w in zip(got, want):
+        torch.testing.assert_close(g, w, rtol=1e-5, atol=1e-6)
+
+
+def test_param_broadcast_from_rank0(tmp_path):
+    world = 2
+    rdv = tmp_path / 'rdv_bcast'
+    mp.spawn(_worker, nprocs=world,
+             args=(world, str(rdv), 1.0, str(tmp_path)), join=True)
+    saved = torch.load(tmp_path / 'params.pt', weights_only=False)
+    ref = _build_net(seed=0)  # rank 0 used seed 0
+    # compare parameters only: BN running stats changed during the forward
+    for k, v in ref.named_parameters():
+        torch.testing.assert_close(saved[k], v)
+
+
+def test_no_sync_skips_communication(tmp_path):
+    world = 2
+    rdv = tmp_path / 'rdv_nosync'
+    mp.spawn(_worker_no_sync, nprocs=world,
+             args=(world, str(rdv), str(tmp_path)), join=True)
+    got = torch.load(tmp_path / 'grads_nosync.pt', weights_only=False)
+    # rank 1's grads must be its LOCAL grads (unsynced)
+    net = _build_net()
+    net(_rank_input(1)).sum().backward()
+    for g, w in zip(got, [p.grad for p in net.parameters()]):
+        torch.testing.assert_close(g, w)
+
+
+def test_bucket_partitioning():
+    from real_time_helmet_detection_amd.parallel.ddp import \
+        BucketedDataParallel
+    net = _build_net()
+    # dist not initialized -> world_size 1, but buckets still built
+    ddp = BucketedDataParallel(net, bucket_cap_mb=0.125,
+                               broadcast_params=False)
+    n_params = sum(1 for p in net.parameters() if p.requires_grad)
+    assert sum(len(b.params) for b in ddp.buckets) == n_params
+    total = sum(p.numel() for p in net.parameters() if p.requires_grad)
+    assert sum(b.numel for b in ddp.buckets) == total
+    assert len(ddp.buckets) > 1  # small cap => multiple buckets
+    cap_elems = 0.125 * 1024 * 1024 / 4
+    big = [b for b in ddp.buckets if b.numel > cap_elems]
+    # only buckets containing a single oversized param may exceed the cap
+    assert all(len(b.params) == 1 for b in big)

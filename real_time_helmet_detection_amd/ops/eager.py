@@ -104,14 +104,14 @@ def batched_decode(heatmap, offset, wh, scale_factor, topk, pool_size,
     pooled = F.max_pool2d(heatmap, pool_size, stride=1, padding=pad)
     peakmap = heatmap * (pooled == heatmap)
 
-    scores, indices = peakmap.view(b, -1).topk(topk, dim=1)
+    scores, indices = peakmap.reshape(b, -1).topk(topk, dim=1)
     clss = torch.div(indices, hw, rounding_mode='floor')
     inds = torch.remainder(indices, hw)
     yinds = torch.div(inds, w, rounding_mode='floor')
     xinds = torch.remainder(inds, w)
 
-    flat_off = offset.view(b, 2, hw)
-    flat_wh = wh.view(b, 2, hw)
+    flat_off = offset.reshape(b, 2, hw)
+    flat_wh = wh.reshape(b, 2, hw)
     xoffs = flat_off[:, 0].gather(1, inds)
     yoffs = flat_off[:, 1].gather(1, inds)
     xsizs = flat_wh[:, 0].gather(1, inds)

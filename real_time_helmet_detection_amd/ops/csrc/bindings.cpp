@@ -1,0 +1,85 @@
+// pybind bindings for the gfx950 kernel set (real_time_helmet_detection_amd.ops._C)
+#include <torch/extension.h>
+
+namespace rthd {
+torch::Tensor add_act_fwd(torch::Tensor a, torch::Tensor b, int64_t act);
+torch::Tensor add_act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act);
+
+std::vector<torch::Tensor> pool2x2_fwd(torch::Tensor x, bool is_max,
+                                       bool need_arg);
+torch::Tensor pool2x2_bwd(torch::Tensor dy, torch::Tensor arg, bool is_max,
+                          int64_t H, int64_t W);
+std::vector<torch::Tensor> maxpool_same_fwd(torch::Tensor x, int64_t k,
+                                            bool need_arg);
+torch::Tensor maxpool_same_bwd(torch::Tensor dy, torch::Tensor arg,
+                               int64_t k);
+
+torch::Tensor upsample2x_add_fwd(torch::Tensor x,
+                                 c10::optional<torch::Tensor> skip);
+torch::Tensor upsample2x_bwd(torch::Tensor dy);
+
+std::vector<torch::Tensor> centernet_loss_fwd(
+    torch::Tensor phm, torch::Tensor poff, torch::Tensor psize,
+    torch::Tensor ghm, torch::Tensor goff, torch::Tensor gsize,
+    torch::Tensor mask, double alpha, double beta);
+std::vector<torch::Tensor> centernet_loss_bwd(
+    torch::Tensor phm, torch::Tensor poff, torch::Tensor psize,
+    torch::Tensor ghm, torch::Tensor goff, torch::Tensor gsize,
+    torch::Tensor mask, torch::Tensor sums, torch::Tensor gout,
+    double alpha, double beta);
+
+std::vector<torch::Tensor> decode_fwd(torch::Tensor hm, torch::Tensor off,
+                                      torch::Tensor wh, int64_t scale_factor,
+                                      int64_t topk, int64_t pool_size,
+                                      bool normalized);
+torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
+                      double iou_threshold);
+
+torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16);
+torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
+                       torch::Tensor scale, torch::Tensor shift,
+                       c10::optional<torch::Tensor> skip,
+                       int64_t KH, int64_t KW, int64_t stride, int64_t pad,
+                       int64_t Cout, int64_t act);
+torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
+                       torch::Tensor scale, torch::Tensor shift,
+                       int64_t stride, int64_t pad, int64_t act);
+torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
+                    int64_t KW, int64_t stride, int64_t pad);
+
+std::vector<torch::Tensor> bn_stats(torch::Tensor x,
+                                    c10::optional<torch::Tensor> running_mean,
+                                    c10::optional<torch::Tensor> running_var,
+                                    double momentum, double eps);
+torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
+                         torch::Tensor rstd, torch::Tensor gamma,
+                         torch::Tensor beta, int64_t act);
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                      torch::Tensor mean, torch::Tensor rstd,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      int64_t act);
+torch::Tensor col_sum(torch::Tensor x);
+}  // namespace rthd
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("add_act_fwd", &rthd::add_act_fwd);
+  m.def("add_act_bwd", &rthd::add_act_bwd);
+  m.def("pool2x2_fwd", &rthd::pool2x2_fwd);
+  m.def("pool2x2_bwd", &rthd::pool2x2_bwd);
+  m.def("maxpool_same_fwd", &rthd::maxpool_same_fwd);
+  m.def("maxpool_same_bwd", &rthd::maxpool_same_bwd);
+  m.def("upsample2x_add_fwd", &rthd::upsample2x_add_fwd);
+  m.def("upsample2x_bwd", &rthd::upsample2x_bwd);
+  m.def("centernet_loss_fwd", &rthd::centernet_loss_fwd);
+  m.def("centernet_loss_bwd", &rthd::centernet_loss_bwd);
+  m.def("decode_fwd", &rthd::decode_fwd);
+  m.def("nms_fwd", &rthd::nms_fwd);
+  m.def("pack_weights", &rthd::pack_weights);
+  m.def("conv_fwd", &rthd::conv_fwd);
+  m.def("stem_fwd", &rthd::stem_fwd);
+  m.def("wgrad", &rthd::wgrad);
+  m.def("bn_stats", &rthd::bn_stats);
+  m.def("bn_act_fwd", &rthd::bn_act_fwd);
+  m.def("bn_act_bwd", &rthd::bn_act_bwd);
+  m.def("col_sum", &rthd::col_sum);
+}

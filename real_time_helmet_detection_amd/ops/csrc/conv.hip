@@ -1,0 +1,453 @@
+// Implicit-GEMM convolution for gfx950 MFMA, NHWC, with fused
+// scale/shift(+bias)/activation(+residual-add) epilogue.
+//
+// GEMM view (guide §5 anatomy): M = B*Ho*Wo output pixels, N = Cout,
+// K = KH*KW*Cin, tap-major. Each workgroup computes a BM=128 x BN=128
+// output tile with 4 waves (2x2 of 64x64 wave tiles, 4x4 fragments of
+// v_mfma_f32_16x16x32_bf16 / _16x16x4_f32). The K loop walks taps x
+// 32-channel blocks; for each step the A tile (128 px x 32 ch) is gathered
+// with zero-padding predication and the B tile (128 cout x 32 ch) is read
+// from the pre-packed weight buffer, both staged in double-buffered LDS
+// with an XOR slot swizzle (guide §6 G4) to keep ds_read_b128 conflict-low.
+//
+// Weights are pre-packed by pack_weights_* into [taps][Cout_pad][Cin_pad]
+// (k contiguous per output-channel row) so A and B fragments use the SAME
+// LDS image and read pattern. dgrad reuses this kernel with rotated,
+// transposed packed weights (pack is done by the python wrapper calling
+// pack_weights with swap=true).
+//
+// Requires Cin % 32 == 0 (every conv in the model except the 3-channel
+// stem, which has its own kernel in stem.hip).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace rthd {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// ------------------------------ weight packing ------------------------------
+
+// torch weight (Cout, Cin, KH, KW) fp32 -> packed [T][Coutp][Cinp] (T dtype)
+// swap=false: pk[t][co][ci] = w[co][ci][t/KW][t%KW]
+// swap=true (dgrad): roles swapped + taps rotated:
+//   pk[t][ci][co] = w[co][ci][KH-1-t/KW][KW-1-t%KW]  (rows indexed by ci)
+template <typename T>
+__global__ void pack_weights_kernel(const float* __restrict__ w,
+                                    T* __restrict__ pk,
+                                    int Cout, int Cin, int KH, int KW,
+                                    int Rows, int Rp, int Kp, int swap) {
+  // Rows/Rp: row count (+pad) of pk (= Cout or Cin); Kp: padded k per tap
+  const int T_ = KH * KW;
+  const int64_t n = (int64_t)T_ * Rp * Kp;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int t = i / ((int64_t)Rp * Kp);
+    const int row = (i / Kp) % Rp;
+    const int k = i % Kp;
+    float v = 0.f;
+    const int Kdim = swap ? Cout : Cin;
+    if (row < Rows && k < Kdim) {
+      int co, ci, ty, tx;
+      if (!swap) {
+        co = row; ci = k; ty = t / KW; tx = t % KW;
+      } else {
+        ci = row; co = k; ty = KH - 1 - t / KW; tx = KW - 1 - t % KW;
+      }
+      v = w[(((int64_t)co * Cin + ci) * KH + ty) * KW + tx];
+    }
+    stf(&pk[i], v);
+  }
+}
+
+torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16) {
+  auto wc = w.to(at::kFloat).contiguous();  // (Cout, Cin, KH, KW)
+  const int Cout = wc.size(0), Cin = wc.size(1);
+  const int KH = wc.size(2), KW = wc.size(3);
+  const int T_ = KH * KW;
+  const int Rows = swap ? Cin : Cout;
+  const int Kdim = swap ? Cout : Cin;
+  const int Rp = (int)cdiv(Rows, 128) * 128;
+  const int Kp = (int)cdiv(Kdim, 32) * 32;
+  auto opt = wc.options().dtype(to_bf16 ? at::kBFloat16 : at::kFloat);
+  auto pk = torch::empty({T_, Rp, Kp}, opt);
+  const int64_t n = (int64_t)T_ * Rp * Kp;
+  auto s = at::cuda::getCurrentCUDAStream();
+  if (to_bf16)
+    hipLaunchKernelGGL((pack_weights_kernel<bf16>), dim3(ew_grid(n, 256)),
+        dim3(256), 0, s, wc.data_ptr<float>(),
+        reinterpret_cast<bf16*>(pk.data_ptr()), Cout, Cin, KH, KW,
+        Rows, Rp, Kp, swap ? 1 : 0);
+  else
+    hipLaunchKernelGGL((pack_weights_kernel<float>), dim3(ew_grid(n, 256)),
+        dim3(256), 0, s, wc.data_ptr<float>(), pk.data_ptr<float>(),
+        Cout, Cin, KH, KW, Rows, Rp, Kp, swap ? 1 : 0);
+  HIP_CHECK_LAST();
+  return pk;
+}
+
+// ------------------------------ conv forward --------------------------------
+
+// LDS tile: 128 rows x 32 k of bf16 (64 B rows); slot swizzle:
+//   byte(row, k8) = row*64 + ((k8 ^ ((row>>2)&3))*16)
+// fp32: 128-B rows, element swizzle k' = k ^ (row & 15) within the row.
+DEV_INLINE int lds_off_bf16(int row, int k8) {
+  return row * 64 + ((k8 ^ ((row >> 2) & 3)) << 4);
+}
+DEV_INLINE int lds_off_f32(int row, int k) {
+  return row * 128 + ((k ^ (row & 15)) << 2);
+}
+
+struct ConvGeo {
+  int B, H, W, Cin, Ho, Wo, Cout;
+  int KH, KW, stride, pad;
+  int Cinp;   // padded Cin (mult of 32)
+  int Coutp;  // padded Cout (mult of 128)
+  int M;      // B*Ho*Wo
+};
+
+// act codes from common.h; epilogue: y = act(acc*scale[c] + shift[c] (+skip))
+template <bool HAS_SKIP>
+__global__ __launch_bounds__(256)
+void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
+                          const bf16* __restrict__ wpk,
+                          const float* __restrict__ scale,
+                          const float* __restrict__ shift,
+                          const bf16* __restrict__ skip,
+                          bf16* __restrict__ y,
+                          ConvGeo g, int act) {
+  // grid: (M/128) x (Coutp/128)
+  const int mblk = blockIdx.x;
+  const int nblk = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 4 waves: 2x2
+  const int wr = wid >> 1, wc = wid & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* ldsA = reinterpret_cast<bf16*>(smem);             // 2 x 8 KB
+  bf16* ldsB = reinterpret_cast<bf16*>(smem + 16384);     // 2 x 8 KB
+
+  f32x4 acc[4][4] = {};
+
+  // staging indices: 256 threads x 16 B = 4 KB per pass; A tile 8 KB -> 2
+  // passes; each thread covers rows (tid/4) and (tid/4 + 64), k8 = tid%4.
+  const int st_row = tid >> 2;
+  const int st_k8 = tid & 3;
+
+  // per-thread A source pixel decomposition for its two staging rows
+  int am[2], ab[2], ay[2], ax[2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int m = mblk * 128 + st_row + 64 * h;
+    am[h] = m;
+    const int mm = m < g.M ? m : 0;
+    ab[h] = mm / (g.Ho * g.Wo);
+    const int r = mm % (g.Ho * g.Wo);
+    ay[h] = r / g.Wo;
+    ax[h] = r % g.Wo;
+  }
+
+  const int nsteps = g.KH * g.KW * (g.Cinp / 32);
+  const int kc_per_tap = g.Cinp / 32;
+
+  for (int step = 0; step < nsteps; ++step) {
+    const int t = step / kc_per_tap;
+    const int kb = step % kc_per_tap;
+    const int dy = t / g.KW - g.pad;
+    const int dx = t % g.KW - g.pad;
+    const int buf = step & 1;
+    bf16* A = ldsA + buf * 4096;
+    bf16* B = ldsB + buf * 4096;
+
+    // ---- stage A (2 rows per thread) ----
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int iy = ay[h] * g.stride + dy;
+      const int ix = ax[h] * g.stride + dx;
+      const int c0 = kb * 32 + st_k8 * 8;
+      uint4 v = {0, 0, 0, 0};
+      if (am[h] < g.M && iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
+          c0 < g.Cin) {
+        v = *reinterpret_cast<const uint4*>(
+            x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0);
+      }
+      *reinterpret_cast<uint4*>(
+          reinterpret_cast<char*>(A) +
+          lds_off_bf16(st_row + 64 * h, st_k8)) = v;
+    }
+
+    // ---- stage B (2 rows per thread) ----
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int row = st_row + 64 * h;  // cout within tile
+      const int64_t src_off =
+          ((int64_t)t * g.Coutp + nblk * 128 + row) * g.Cinp + kb * 32 +
+          st_k8 * 8;
+      const uint4 v = *reinterpret_cast<const uint4*>(wpk + src_off);
+      *reinterpret_cast<uint4*>(
+          reinterpret_cast<char*>(B) + lds_off_bf16(row, st_k8)) = v;
+    }
+
+    __syncthreads();
+
+    // ---- fragments + MFMA ----
+    const int arow_base = wr * 64 + (lane & 15);
+    const int brow_base = wc * 64 + (lane & 15);
+    const int k8 = lane >> 4;
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      afrag[i] = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<char*>(A) + lds_off_bf16(arow_base + 16 * i, k8));
+      bfrag[i] = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<char*>(B) + lds_off_bf16(brow_base + 16 * i, k8));
+    }
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+
+    __syncthreads();
+  }
+
+  // ---- epilogue ----
+  const int col0 = nblk * 128 + wc * 64 + (lane & 15);
+  const int row_in_frag = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mblk * 128 + wr * 64 + mi * 16 + row_in_frag + r;
+      if (m >= g.M) continue;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int c = col0 + ni * 16;
+        if (c >= g.Cout) continue;
+        float v = acc[mi][ni][r];
+        v = v * scale[c] + shift[c];
+        if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
+        v = apply_act(v, act);
+        stf(&y[(int64_t)m * g.Cout + c], v);
+      }
+    }
+  }
+}
+
+// fp32-exact variant: v_mfma_f32_16x16x4_f32, one A/B float per lane
+template <bool HAS_SKIP>
+__global__ __launch_bounds__(256)
+void conv_fwd_f32_kernel(const float* __restrict__ x,
+                         const float* __restrict__ wpk,
+                         const float* __restrict__ scale,
+                         const float* __restrict__ shift,
+                         const float* __restrict__ skip,
+                         float* __restrict__ y,
+                         ConvGeo g, int act) {
+  const int mblk = blockIdx.x;
+  const int nblk = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* ldsA = reinterpret_cast<float*>(smem);            // 16 KB
+  float* ldsB = reinterpret_cast<float*>(smem + 16384);    // 16 KB
+
+  f32x4 acc[4][4] = {};
+
+  const int st_row = tid >> 2;   // 2 rows per thread per tile
+  const int st_k4 = tid & 3;     // 4 floats (16 B) x ... k chunk of 8? no:
+  // A tile rows are 32 floats (128 B): 8 chunks of 16 B -> use 2 passes of
+  // 4-chunk? Simplest: each thread stages 2 rows x 8 floats:
+  const int st_k8 = (tid & 3) * 8;
+
+  int am[2], ab[2], ayy[2], axx[2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int m = mblk * 128 + st_row + 64 * h;
+    am[h] = m;
+    const int mm = m < g.M ? m : 0;
+    ab[h] = mm / (g.Ho * g.Wo);
+    const int r = mm % (g.Ho * g.Wo);
+    ayy[h] = r / g.Wo;
+    axx[h] = r % g.Wo;
+  }
+
+  const int nsteps = g.KH * g.KW * (g.Cinp / 32);
+  const int kc_per_tap = g.Cinp / 32;
+
+  for (int step = 0; step < nsteps; ++step) {
+    const int t = step / kc_per_tap;
+    const int kb = step % kc_per_tap;
+    const int dy = t / g.KW - g.pad;
+    const int dx = t % g.KW - g.pad;
+    float* A = ldsA;
+    float* B = ldsB;
+    __syncthreads();  // previous step's reads done before overwrite
+
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int iy = ayy[h] * g.stride + dy;
+      const int ix = axx[h] * g.stride + dx;
+      const int c0 = kb * 32 + st_k8;
+      float v[8] = {};
+      if (am[h] < g.M && iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
+          c0 < g.Cin) {
+        const float* src =
+            x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = src[e];
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        *reinterpret_cast<float*>(
+            reinterpret_cast<char*>(A) +
+            lds_off_f32(st_row + 64 * h, st_k8 + e)) = v[e];
+      }
+    }
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int row = st_row + 64 * h;
+      const int64_t src_off =
+          ((int64_t)t * g.Coutp + nblk * 128 + row) * g.Cinp + kb * 32 +
+          st_k8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        *reinterpret_cast<float*>(
+            reinterpret_cast<char*>(B) + lds_off_f32(row, st_k8 + e)) =
+            wpk[src_off + e];
+      }
+    }
+
+    __syncthreads();
+
+    const int arow = wr * 64 + (lane & 15);
+    const int brow = wc * 64 + (lane & 15);
+    const int kl = lane >> 4;  // 0..3
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks) {  // 8 x K=4 = 32
+      float afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const float*>(
+            reinterpret_cast<char*>(A) +
+            lds_off_f32(arow + 16 * i, ks * 4 + kl));
+        bfrag[i] = *reinterpret_cast<const float*>(
+            reinterpret_cast<char*>(B) +
+            lds_off_f32(brow + 16 * i, ks * 4 + kl));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+
+    __syncthreads();
+  }
+
+  const int col0 = nblk * 128 + wc * 64 + (lane & 15);
+  const int row_in_frag = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mblk * 128 + wr * 64 + mi * 16 + row_in_frag + r;
+      if (m >= g.M) continue;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int c = col0 + ni * 16;
+        if (c >= g.Cout) continue;
+        float v = acc[mi][ni][r];
+        v = v * scale[c] + shift[c];
+        if (HAS_SKIP) v += skip[(int64_t)m * g.Cout + c];
+        v = apply_act(v, act);
+        y[(int64_t)m * g.Cout + c] = v;
+      }
+    }
+  }
+}
+
+// host wrapper; x NCHW-logical channels_last; wpk from pack_weights.
+torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
+                       torch::Tensor scale, torch::Tensor shift,
+                       c10::optional<torch::Tensor> skip,
+                       int64_t KH, int64_t KW, int64_t stride, int64_t pad,
+                       int64_t Cout, int64_t act) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  ConvGeo g;
+  g.B = xc.size(0);
+  g.Cin = xc.size(1);
+  g.H = xc.size(2);
+  g.W = xc.size(3);
+  g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
+  g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
+  g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
+  g.Cout = Cout;
+  g.Cinp = (int)cdiv(g.Cin, 32) * 32;
+  g.Coutp = (int)cdiv(Cout, 128) * 128;
+  g.M = g.B * g.Ho * g.Wo;
+  TORCH_CHECK(g.Cin % 32 == 0, "conv_fwd requires Cin % 32 == 0 (got ",
+              g.Cin, ") — the 3-channel stem uses stem_fwd");
+  TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == g.Coutp &&
+              wpk.size(2) == g.Cinp, "conv_fwd: packed weight shape");
+
+  const bool bf16_mode = wpk.scalar_type() == at::kBFloat16;
+  auto out_dtype = bf16_mode ? at::kBFloat16 : at::kFloat;
+  auto y = torch::empty({g.B, (int64_t)g.Cout, g.Ho, g.Wo},
+                        xc.options().dtype(out_dtype)
+                        .memory_format(at::MemoryFormat::ChannelsLast));
+  auto sc = scale.to(at::kFloat).contiguous();
+  auto sh = shift.to(at::kFloat).contiguous();
+
+  const bool has_skip = skip.has_value();
+  torch::Tensor sk;
+  if (has_skip)
+    sk = skip->to(out_dtype).contiguous(at::MemoryFormat::ChannelsLast);
+
+  dim3 grid(cdiv(g.M, 128), g.Coutp / 128);
+  auto s = at::cuda::getCurrentCUDAStream();
+
+  if (bf16_mode) {
+    const size_t lds = 32768;
+    if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
+    auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
+    auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());
+    auto* py = reinterpret_cast<bf16*>(y.data_ptr());
+    const bf16* ps =
+        has_skip ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
+    if (has_skip)
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true>), grid,
+          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
+          sh.data_ptr<float>(), ps, py, g, (int)act);
+    else
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false>), grid,
+          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
+          sh.data_ptr<float>(), ps, py, g, (int)act);
+  } else {
+    const size_t lds = 32768;
+    TORCH_CHECK(xc.scalar_type() == at::kFloat, "f32 conv needs f32 input");
+    const float* ps = has_skip ? sk.data_ptr<float>() : nullptr;
+    if (has_skip)
+      hipLaunchKernelGGL((conv_fwd_f32_kernel<true>), grid, dim3(256), lds,
+          s, xc.data_ptr<float>(), wpk.data_ptr<float>(),
+          sc.data_ptr<float>(), sh.data_ptr<float>(), ps,
+          y.data_ptr<float>(), g, (int)act);
+    else
+      hipLaunchKernelGGL((conv_fwd_f32_kernel<false>), grid, dim3(256), lds,
+          s, xc.data_ptr<float>(), wpk.data_ptr<float>(),
+          sc.data_ptr<float>(), sh.data_ptr<float>(), ps,
+          y.data_ptr<float>(), g, (int)act);
+  }
+  HIP_CHECK_LAST();
+  return y;
+}
+
+}  // namespace rthd

@@ -1,0 +1,130 @@
+// Greedy IoU NMS, single-workgroup LDS-resident (boxes after decode +
+// confidence threshold are few hundred at most: S*topk <= 2048).
+//
+// Replaces torchvision.ops.nms (reference evaluate.py:174) and the scripted
+// export NMS: sort by score (bitonic, desc, index-stable) entirely in LDS,
+// then the greedy suppression scan with all lanes testing IoU in parallel.
+// Returns kept indices (original numbering) in descending-score order.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace rthd {
+
+constexpr int NMS_CAP = 2048;
+
+__global__ void nms_kernel(const float* __restrict__ boxes,  // (N,4)
+                           const float* __restrict__ scores, // (N,)
+                           int* __restrict__ out_idx,        // (N,)
+                           int* __restrict__ out_count,
+                           int N, float thr) {
+  __shared__ float sx1[NMS_CAP], sy1[NMS_CAP], sx2[NMS_CAP], sy2[NMS_CAP];
+  __shared__ float ss[NMS_CAP];
+  __shared__ int sidx[NMS_CAP];
+  __shared__ unsigned char kept[NMS_CAP];
+
+  const int P = NMS_CAP;  // sort width (pow2)
+  for (int i = threadIdx.x; i < P; i += blockDim.x) {
+    if (i < N) {
+      sx1[i] = boxes[i * 4 + 0];
+      sy1[i] = boxes[i * 4 + 1];
+      sx2[i] = boxes[i * 4 + 2];
+      sy2[i] = boxes[i * 4 + 3];
+      ss[i] = scores[i];
+      sidx[i] = i;
+    } else {
+      ss[i] = -3.4e38f;
+      sidx[i] = 0x7fffffff;
+    }
+    kept[i] = 1;
+  }
+  __syncthreads();
+
+  // bitonic sort desc by (score, -idx)
+  for (int k2 = 2; k2 <= P; k2 <<= 1) {
+    for (int j2 = k2 >> 1; j2 > 0; j2 >>= 1) {
+      for (int i = threadIdx.x; i < P / 2; i += blockDim.x) {
+        const int a = (i / j2) * (j2 * 2) + (i % j2);
+        const int b = a ^ j2;
+        if (b > a) {
+          const bool dirDesc = ((a & k2) == 0);
+          const bool a_lt_b = (ss[a] < ss[b]) ||
+              (ss[a] == ss[b] && sidx[a] > sidx[b]);
+          if (dirDesc == a_lt_b) {
+            float t;
+            int ti;
+            t = ss[a]; ss[a] = ss[b]; ss[b] = t;
+            ti = sidx[a]; sidx[a] = sidx[b]; sidx[b] = ti;
+            if (sidx[a] < N || sidx[b] < N) {
+              t = sx1[a]; sx1[a] = sx1[b]; sx1[b] = t;
+              t = sy1[a]; sy1[a] = sy1[b]; sy1[b] = t;
+              t = sx2[a]; sx2[a] = sx2[b]; sx2[b] = t;
+              t = sy2[a]; sy2[a] = sy2[b]; sy2[b] = t;
+            }
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // greedy suppression
+  for (int i = 0; i < N; ++i) {
+    if (kept[i]) {
+      const float ax1 = sx1[i], ay1 = sy1[i], ax2 = sx2[i], ay2 = sy2[i];
+      const float area_a = fmaxf(ax2 - ax1, 0.f) * fmaxf(ay2 - ay1, 0.f);
+      for (int j = i + 1 + threadIdx.x; j < N; j += blockDim.x) {
+        if (!kept[j]) continue;
+        const float xx1 = fmaxf(ax1, sx1[j]);
+        const float yy1 = fmaxf(ay1, sy1[j]);
+        const float xx2 = fminf(ax2, sx2[j]);
+        const float yy2 = fminf(ay2, sy2[j]);
+        const float iw = fmaxf(xx2 - xx1, 0.f);
+        const float ih = fmaxf(yy2 - yy1, 0.f);
+        const float inter = iw * ih;
+        const float area_b =
+            fmaxf(sx2[j] - sx1[j], 0.f) * fmaxf(sy2[j] - sy1[j], 0.f);
+        const float uni = area_a + area_b - inter;
+        const float iou = inter / fmaxf(uni, 1e-9f);
+        if (iou > thr) kept[j] = 0;
+      }
+    }
+    __syncthreads();
+  }
+
+  // emit kept (already in desc-score order)
+  __shared__ int cnt;
+  if (threadIdx.x == 0) cnt = 0;
+  __syncthreads();
+  // ordered compaction by a single wave scan to keep output sorted
+  if (threadIdx.x == 0) {
+    int k = 0;
+    for (int i = 0; i < N; ++i)
+      if (kept[i]) out_idx[k++] = sidx[i];
+    *out_count = k;
+    cnt = k;
+  }
+}
+
+torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
+                      double iou_threshold) {
+  auto b = boxes.to(at::kFloat).contiguous();
+  auto sc = scores.to(at::kFloat).contiguous();
+  const int N = b.size(0);
+  TORCH_CHECK(N <= NMS_CAP,
+              "nms: too many boxes (", N, " > ", NMS_CAP,
+              ") — raise conf_th or topk split");
+  auto out_idx = torch::empty({std::max(N, 1)},
+                              b.options().dtype(at::kInt));
+  auto out_count = torch::zeros({1}, b.options().dtype(at::kInt));
+  if (N == 0) return torch::empty({0}, b.options().dtype(at::kLong));
+  auto s = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(nms_kernel, dim3(1), dim3(256), 0, s,
+      b.data_ptr<float>(), sc.data_ptr<float>(), out_idx.data_ptr<int>(),
+      out_count.data_ptr<int>(), N, (float)iou_threshold);
+  HIP_CHECK_LAST();
+  const int k = out_count.item<int>();  // host sync: result used on host
+  return out_idx.narrow(0, 0, k).to(at::kLong);
+}
+
+}  // namespace rthd

@@ -1,0 +1,108 @@
+// Direct convolution for the 3-channel stem (7x7 stride-2 pad-3, 3->64).
+//
+// Cin=3 defeats the implicit-GEMM K-blocking (conv.hip needs Cin%32==0), and
+// K = 147 taps*ch is tiny, so a direct kernel wins: the full weight set
+// (64 cout x 49 taps x 3 ch fp32 = 37.6 KB) lives in LDS and broadcasts to
+// all lanes; each thread owns ONE output pixel and its full 64-channel
+// output row (64 fp32 accumulators), so the NHWC store is one contiguous
+// 128 B (bf16) row per lane. Epilogue = scale/shift (folded BN or bias) +
+// activation, same contract as conv_fwd.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace rthd {
+
+constexpr int STEM_COUT = 64;
+constexpr int STEM_CIN = 3;
+
+template <typename T, int KS>  // KS = kernel size (7)
+__global__ __launch_bounds__(256)
+void stem_fwd_kernel(const T* __restrict__ x,
+                     const float* __restrict__ w,  // (64, 3, KS, KS) fp32
+                     const float* __restrict__ scale,
+                     const float* __restrict__ shift,
+                     T* __restrict__ y,
+                     int B, int H, int W, int Ho, int Wo,
+                     int stride, int pad, int act) {
+  __shared__ float wl[KS * KS * STEM_CIN * STEM_COUT];  // [t][ci][co]
+  // repack w (co, ci, ty, tx) -> wl[t*3*64 + ci*64 + co] for broadcast reads
+  for (int i = threadIdx.x; i < KS * KS * STEM_CIN * STEM_COUT;
+       i += blockDim.x) {
+    const int co = i % STEM_COUT;
+    const int ci = (i / STEM_COUT) % STEM_CIN;
+    const int t = i / (STEM_COUT * STEM_CIN);
+    wl[i] = w[((co * STEM_CIN + ci) * KS + t / KS) * KS + t % KS];
+  }
+  __syncthreads();
+
+  const int M = B * Ho * Wo;
+  for (int m = blockIdx.x * blockDim.x + threadIdx.x; m < M;
+       m += gridDim.x * blockDim.x) {
+    const int b = m / (Ho * Wo);
+    const int r = m % (Ho * Wo);
+    const int oy = r / Wo, ox = r % Wo;
+
+    float acc[STEM_COUT] = {};
+#pragma unroll 1
+    for (int ty = 0; ty < KS; ++ty) {
+      const int iy = oy * stride + ty - pad;
+      if (iy < 0 || iy >= H) continue;
+#pragma unroll 1
+      for (int tx = 0; tx < KS; ++tx) {
+        const int ix = ox * stride + tx - pad;
+        if (ix < 0 || ix >= W) continue;
+        const T* px = x + (((int64_t)b * H + iy) * W + ix) * STEM_CIN;
+        const float i0 = ldf(&px[0]), i1 = ldf(&px[1]), i2 = ldf(&px[2]);
+        const float* wt = wl + (ty * KS + tx) * STEM_CIN * STEM_COUT;
+#pragma unroll
+        for (int co = 0; co < STEM_COUT; ++co) {
+          acc[co] = fmaf(i0, wt[co], acc[co]);
+          acc[co] = fmaf(i1, wt[STEM_COUT + co], acc[co]);
+          acc[co] = fmaf(i2, wt[2 * STEM_COUT + co], acc[co]);
+        }
+      }
+    }
+
+    T* out = y + (int64_t)m * STEM_COUT;
+#pragma unroll
+    for (int co = 0; co < STEM_COUT; ++co)
+      stf(&out[co], apply_act(acc[co] * scale[co] + shift[co], act));
+  }
+}
+
+torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
+                       torch::Tensor scale, torch::Tensor shift,
+                       int64_t stride, int64_t pad, int64_t act) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  auto wc = w.to(at::kFloat).contiguous();
+  TORCH_CHECK(wc.size(0) == STEM_COUT && wc.size(1) == STEM_CIN &&
+              wc.size(2) == 7 && wc.size(3) == 7,
+              "stem_fwd supports the 64x3x7x7 stem only");
+  const int B = xc.size(0), H = xc.size(2), W = xc.size(3);
+  const int Ho = (H + 2 * (int)pad - 7) / (int)stride + 1;
+  const int Wo = (W + 2 * (int)pad - 7) / (int)stride + 1;
+  auto y = torch::empty({B, STEM_COUT, Ho, Wo}, xc.options()
+                        .memory_format(at::MemoryFormat::ChannelsLast));
+  auto sc = scale.to(at::kFloat).contiguous();
+  auto sh = shift.to(at::kFloat).contiguous();
+  const int64_t M = (int64_t)B * Ho * Wo;
+  auto s = at::cuda::getCurrentCUDAStream();
+  if (xc.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((stem_fwd_kernel<bf16, 7>), dim3(ew_grid(M, 256)),
+        dim3(256), 0, s, reinterpret_cast<const bf16*>(xc.data_ptr()),
+        wc.data_ptr<float>(), sc.data_ptr<float>(), sh.data_ptr<float>(),
+        reinterpret_cast<bf16*>(y.data_ptr()), B, H, W, Ho, Wo,
+        (int)stride, (int)pad, (int)act);
+  } else {
+    TORCH_CHECK(xc.scalar_type() == at::kFloat);
+    hipLaunchKernelGGL((stem_fwd_kernel<float, 7>), dim3(ew_grid(M, 256)),
+        dim3(256), 0, s, xc.data_ptr<float>(), wc.data_ptr<float>(),
+        sc.data_ptr<float>(), sh.data_ptr<float>(), y.data_ptr<float>(),
+        B, H, W, Ho, Wo, (int)stride, (int)pad, (int)act);
+  }
+  HIP_CHECK_LAST();
+  return y;
+}
+
+}  // namespace rthd

@@ -1,0 +1,324 @@
+"""GPU execution path: autograd wrappers over the gfx950 kernels (_C).
+
+Every function here matches an eager twin in ``ops.eager`` /
+``ops.functional`` — same signatures, same numerics (tested in
+tests/test_gpu_kernels.py against fp32 eager oracles).
+
+Structure per op: a torch.autograd.Function whose forward/backward call the
+HIP kernels directly. Convolutions:
+
+- forward: pack_weights (torch (Cout,Cin,KH,KW) fp32 master -> MFMA-friendly
+  [taps][Coutp][Cinp] bf16/f32) -> conv_fwd implicit GEMM with fused
+  scale/shift/act epilogue. Training-mode BN splits into conv(linear) ->
+  bn_stats -> bn_act_fwd; inference folds BN into the epilogue.
+- backward: dgrad = conv_fwd on dY with rotated/transposed packed weights;
+  wgrad = the transposed-staging MFMA kernel; BN/bias grads from the
+  reduction kernels.
+
+bf16 policy: when rthd.amp autocast is active (or inputs are bf16) compute
+runs on the bf16 MFMA pipe with fp32 accumulate; otherwise the exact-f32
+MFMA path (guide §3: v_mfma_f32_16x16x4_f32 — identical numerics to an fmaf
+chain).
+"""
+
+import torch
+
+from . import _backend
+from .. import amp as _amp
+
+ACT_CODE = {'Linear': 0, 'ReLU': 1, 'LReLU': 2}
+
+
+def _C():
+    return _backend.require_ext()
+
+
+def _bf16_mode(x):
+    return x.dtype == torch.bfloat16 or _amp.is_autocast_enabled()
+
+
+# ------------------------------------------------------------------ conv ---
+
+class _ConvBNActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, gamma, beta, rmean, rvar,
+                kh, kw, stride, pad, act_code, use_bn, training, momentum,
+                eps, is_stem):
+        C = _C()
+        bf16 = _bf16_mode(x)
+        dtype = torch.bfloat16 if bf16 else torch.float32
+        xc = x.to(dtype).contiguous(memory_format=torch.channels_last)
+        cout = weight.shape[0]
+        dev = x.device
+
+        wpk = None
+        if not is_stem:
+            wpk = C.pack_weights(weight, False, bf16)
+
+        ones = torch.ones(cout, device=dev, dtype=torch.float32)
+        zeros = torch.zeros(cout, device=dev, dtype=torch.float32)
+
+        def run_conv(scale, shift, act):
+            if is_stem:
+                return C.stem_fwd(xc, weight, scale, shift, stride, pad, act)
+            return C.conv_fwd(xc, wpk, scale, shift, None, kh, kw, stride,
+                              pad, cout, act)
+
+        bias_f = bias.float().contiguous() if bias is not None else zeros
+        mean = rstd = y_lin = None
+        if use_bn and training:
+            # conv output INCLUDES the (redundant-under-BN) bias so running
+            # stats match the eager/reference semantics (stem has bias+BN)
+            y_lin = run_conv(ones, bias_f, ACT_CODE['Linear'])
+            mean, rstd = C.bn_stats(y_lin, rmean, rvar, momentum, eps)
+            y = C.bn_act_fwd(y_lin, mean, rstd, gamma, beta, act_code)
+        elif use_bn:
+            rstd_run = torch.rsqrt(rvar.float() + eps)
+            scale = (gamma.float() * rstd_run).contiguous()
+            shift = (beta.float() + (bias_f - rmean.float()) * scale
+                     ).contiguous()
+            y = run_conv(scale, shift, act_code)
+        else:
+            y = run_conv(ones, bias_f, act_code)
+
+        ctx.meta = (kh, kw, stride, pad, act_code, use_bn, training, bf16,
+                    is_stem, bias is not None, eps)
+        if use_bn and training:
+            ctx.save_for_backward(xc, weight, gamma, beta, y_lin, mean, rstd)
+        elif use_bn:
+            ctx.save_for_backward(xc, weight, gamma, beta, y,
+                                  rmean.detach().clone(),
+                                  rvar.detach().clone())
+        else:
+            ctx.save_for_backward(xc, weight, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _C()
+        (kh, kw, stride, pad, act_code, use_bn, training, bf16, is_stem,
+         has_bias, eps) = ctx.meta
+        dgamma = dbeta = dbias = None
+
+        if use_bn and training:
+            xc, weight, gamma, beta, y_lin, mean, rstd = ctx.saved_tensors
+            dpre, dgamma, dbeta = C.bn_act_bwd(dy, y_lin, mean, rstd,
+                                               gamma, beta, act_code)
+        elif use_bn:
+            # eval-mode BN backward (rare: grads through a frozen BN)
+            xc, weight, gamma, beta, y, rmean, rvar = ctx.saved_tensors
+            dact = C.add_act_bwd(dy, y, act_code)
+            scale = (gamma.float() * torch.rsqrt(rvar.float() + eps))
+            dpre = (dact.float() * scale.view(1, -1, 1, 1)).to(dact.dtype)
+            dpre = dpre.contiguous(memory_format=torch.channels_last)
+        else:
+            xc, weight, y = ctx.saved_tensors
+            dpre = C.add_act_bwd(dy, y, act_code)
+        if has_bias and ctx.needs_input_grad[2]:
+            dbias = C.col_sum(dpre)
+
+        # dgrad
+        dx = None
+        if ctx.needs_input_grad[0]:
+            if is_stem:
+                raise NotImplementedError(
+                    'stem dgrad (3-channel input) is not needed: the image '
+                    'is a leaf tensor')
+            if stride != 1:
+                raise NotImplementedError(
+                    'dgrad for stride>1 convs (pool="Conv") is not '
+                    'implemented in the HIP engine yet')
+            cin = xc.shape[1]
+            wpk_t = C.pack_weights(weight, True, bf16)
+            ones = torch.ones(cin, device=xc.device, dtype=torch.float32)
+            zeros = torch.zeros(cin, device=xc.device, dtype=torch.float32)
+            dx = C.conv_fwd(dpre, wpk_t, ones, zeros, None, kh, kw, 1, pad,
+                            cin, ACT_CODE['Linear'])
+
+        # wgrad
+        dw = None
+        if ctx.needs_input_grad[1]:
+            dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
+
+        return (dx, dw, dbias, dgamma, dbeta, None, None,
+                None, None, None, None, None, None, None, None, None, None)
+
+
+def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
+    """GPU twin of functional.conv_bn_act (act_module path stays eager)."""
+    kh, kw = conv.kernel_size
+    stride = conv.stride[0]
+    pad = conv.padding[0]
+    cin = conv.weight.shape[1]
+    is_stem = cin < 32
+    if is_stem:
+        if (conv.weight.shape[0] != 64 or cin != 3 or kh != 7 or kw != 7):
+            raise NotImplementedError(
+                'HIP engine: small-Cin convs support only the 64x3x7x7 stem '
+                f'(got {tuple(conv.weight.shape)})')
+    use_bn = bn is not None
+    act_code = ACT_CODE.get(act)
+    if act_code is None:
+        raise NotImplementedError(f'HIP conv epilogue: activation {act!r}')
+    y = _ConvBNActFn.apply(
+        x, conv.weight,
+        conv.bias,
+        bn.weight if use_bn else None,
+        bn.bias if use_bn else None,
+        bn.running_mean if use_bn else None,
+        bn.running_var if use_bn else None,
+        kh, kw, stride, pad, act_code, use_bn, training,
+        bn.momentum if use_bn else 0.1,
+        bn.eps if use_bn else 1e-5,
+        is_stem)
+    if act_module is not None:
+        y = act_module(y)
+    return y
+
+
+# ------------------------------------------------------------ elementwise --
+
+class _AddActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, act_code):
+        y = _C().add_act_fwd(a, b, act_code)
+        ctx.act_code = act_code
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dz = _C().add_act_bwd(dy, y, ctx.act_code)
+        return dz, dz, None
+
+
+def add_act(a, b, act='Linear', act_module=None):
+    code = ACT_CODE.get(act)
+    if code is None:
+        raise NotImplementedError(f'HIP add_act: activation {act!r}')
+    y = _AddActFn.apply(a, b, code)
+    if act_module is not None:
+        y = act_module(y)
+    return y
+
+
+# ----------------------------------------------------------------- pools ---
+
+class _MaxPool2x2Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        need_arg = x.requires_grad
+        out = _C().pool2x2_fwd(x, True, need_arg)
+        if need_arg:
+            ctx.save_for_backward(out[1])
+        ctx.hw = (x.shape[2], x.shape[3])
+        return out[0]
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        return _C().pool2x2_bwd(dy, arg, True, *ctx.hw)
+
+
+class _AvgPool2x2Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.hw = (x.shape[2], x.shape[3])
+        return _C().pool2x2_fwd(x, False, False)[0]
+
+    @staticmethod
+    def backward(ctx, dy):
+        dummy = torch.empty(0, dtype=torch.uint8, device=dy.device)
+        return _C().pool2x2_bwd(dy, dummy, False, *ctx.hw)
+
+
+class _MaxPoolSameFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k):
+        need_arg = x.requires_grad
+        out = _C().maxpool_same_fwd(x, k, need_arg)
+        if need_arg:
+            ctx.save_for_backward(out[1])
+        ctx.k = k
+        return out[0]
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        return _C().maxpool_same_bwd(dy, arg, ctx.k), None
+
+
+def maxpool2x2(x):
+    return _MaxPool2x2Fn.apply(x)
+
+
+def avgpool2x2(x):
+    return _AvgPool2x2Fn.apply(x)
+
+
+def maxpool_same(x, kernel):
+    return _MaxPoolSameFn.apply(x, kernel)
+
+
+# -------------------------------------------------------------- upsample ---
+
+class _Upsample2xAddFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, skip):
+        ctx.has_skip = skip is not None
+        return _C().upsample2x_add_fwd(x, skip)
+
+    @staticmethod
+    def backward(ctx, dy):
+        dx = _C().upsample2x_bwd(dy)
+        return dx, (dy if ctx.has_skip else None)
+
+
+def upsample2x_add(x, skip=None):
+    if skip is None:
+        return _Upsample2xAddFn.apply(x, None)
+    return _Upsample2xAddFn.apply(x, skip)
+
+
+# ------------------------------------------------------------------ loss ---
+
+class _CenterNetLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, phm, poff, psize, ghm, goff, gsize, mask, alpha, beta):
+        losses, sums = _C().centernet_loss_fwd(phm, poff, psize, ghm, goff,
+                                               gsize, mask, alpha, beta)
+        ctx.save_for_backward(phm, poff, psize, ghm, goff, gsize, mask, sums)
+        ctx.ab = (alpha, beta)
+        return losses[0], losses[1], losses[2]
+
+    @staticmethod
+    def backward(ctx, g_hm, g_off, g_size):
+        phm, poff, psize, ghm, goff, gsize, mask, sums = ctx.saved_tensors
+        alpha, beta = ctx.ab
+        gout = torch.stack([g_hm.float(), g_off.float(), g_size.float()])
+        dphm, dpoff, dpsize = _C().centernet_loss_bwd(
+            phm, poff, psize, ghm, goff, gsize, mask, sums, gout, alpha,
+            beta)
+        return (dphm, dpoff, dpsize, None, None, None, None, None, None)
+
+
+def centernet_losses(phm, poff, psize, ghm, goff, gsize, mask,
+                     focal_alpha, focal_beta):
+    return _CenterNetLossFn.apply(phm.float(), poff.float(), psize.float(),
+                                  ghm, goff, gsize, mask,
+                                  float(focal_alpha), float(focal_beta))
+
+
+# ---------------------------------------------------------------- decode ---
+
+def batched_decode(heatmap, offset, wh, scale_factor, topk, pool_size,
+                   normalized):
+    boxes, clss, scores = _C().decode_fwd(heatmap, offset, wh,
+                                          int(scale_factor), int(topk),
+                                          int(pool_size), bool(normalized))
+    return boxes, clss, scores
+
+
+def nms(boxes, scores, iou_threshold):
+    return _C().nms_fwd(boxes, scores, float(iou_threshold))

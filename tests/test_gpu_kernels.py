@@ -1,0 +1,348 @@
+"""HIP kernel parity tests vs fp32 eager oracles (run on MI355X).
+
+Every custom gfx950 kernel is compared against the plain PyTorch fp32
+reference op computed on CPU (SURVEY.md §4 test strategy). bf16 paths use
+relative-norm tolerances sized for bf16 accumulation; fp32 MFMA paths are
+exact-f32 (fmaf-chain numerics) and use tight tolerances.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+CL = torch.channels_last
+
+
+def _C():
+    from real_time_helmet_detection_amd.ops import _backend
+    return _backend.require_ext()
+
+
+def to_gpu(x, dtype=torch.float32):
+    return x.to('cuda', dtype).contiguous(memory_format=CL)
+
+
+def rel_err(got, want):
+    got = got.detach().float().cpu()
+    want = want.detach().float().cpu()
+    denom = want.abs().max().clamp(min=1e-6)
+    return ((got - want).abs().max() / denom).item()
+
+
+# ---------------------------------------------------------------- add_act --
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('act', [0, 1, 2])
+def test_add_act_fwd_bwd(dtype, act):
+    torch.manual_seed(0)
+    a = torch.randn(2, 32, 16, 16)
+    b = torch.randn(2, 32, 16, 16)
+    names = {0: None, 1: 'relu', 2: 'lrelu'}
+    z = a + b
+    want = z if act == 0 else (F.relu(z) if act == 1
+                               else F.leaky_relu(z, 0.01))
+    got = _C().add_act_fwd(to_gpu(a, dtype), to_gpu(b, dtype), act)
+    tol = 0.02 if dtype == torch.bfloat16 else 1e-6
+    assert rel_err(got, want) < tol
+
+    dy = torch.randn_like(z)
+    grad = torch.ones_like(z)
+    if act == 1:
+        grad = (want > 0).float()
+    elif act == 2:
+        grad = torch.where(want > 0, 1.0, 0.01)
+    want_dz = dy * grad
+    got_dz = _C().add_act_bwd(to_gpu(dy, dtype), got, act)
+    assert rel_err(got_dz, want_dz) < (0.02 if dtype == torch.bfloat16
+                                       else 1e-6)
+
+
+# ------------------------------------------------------------------ pools --
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_maxpool2x2(dtype):
+    torch.manual_seed(1)
+    x = torch.randn(2, 32, 16, 16)
+    want = F.max_pool2d(x, 2, 2)
+    out = _C().pool2x2_fwd(to_gpu(x, dtype), True, True)
+    assert rel_err(out[0], want) < (0.01 if dtype == torch.bfloat16 else 1e-6)
+
+    xg = x.clone().requires_grad_(True)
+    dy = torch.randn(2, 32, 8, 8)
+    F.max_pool2d(xg, 2, 2).backward(dy)
+    got_dx = _C().pool2x2_bwd(to_gpu(dy, dtype), out[1], True, 16, 16)
+    assert rel_err(got_dx, xg.grad) < (0.01 if dtype == torch.bfloat16
+                                       else 1e-6)
+
+
+def test_avgpool2x2():
+    torch.manual_seed(2)
+    x = torch.randn(2, 32, 16, 16)
+    want = F.avg_pool2d(x, 2, 2)
+    out = _C().pool2x2_fwd(to_gpu(x), False, False)
+    assert rel_err(out[0], want) < 1e-6
+
+
+@pytest.mark.parametrize('k', [3, 5, 9, 13])
+def test_maxpool_same(k):
+    torch.manual_seed(3)
+    x = torch.randn(2, 32, 16, 16)
+    want = F.max_pool2d(x, k, stride=1, padding=k // 2)
+    out = _C().maxpool_same_fwd(to_gpu(x), k, True)
+    assert rel_err(out[0], want) < 1e-6
+
+    xg = x.clone().requires_grad_(True)
+    dy = torch.randn_like(x)
+    F.max_pool2d(xg, k, stride=1, padding=k // 2).backward(dy)
+    got_dx = _C().maxpool_same_bwd(to_gpu(dy), out[1], k)
+    assert rel_err(got_dx, xg.grad) < 1e-5
+
+
+# --------------------------------------------------------------- upsample --
+
+def test_upsample2x_add():
+    torch.manual_seed(4)
+    x = torch.randn(2, 32, 8, 8)
+    skip = torch.randn(2, 32, 16, 16)
+    want = F.interpolate(x, scale_factor=2, mode='nearest') + skip
+    got = _C().upsample2x_add_fwd(to_gpu(x), to_gpu(skip))
+    assert rel_err(got, want) < 1e-6
+
+    dy = torch.randn(2, 32, 16, 16)
+    xg = x.clone().requires_grad_(True)
+    F.interpolate(xg, scale_factor=2, mode='nearest').backward(dy)
+    got_dx = _C().upsample2x_bwd(to_gpu(dy))
+    assert rel_err(got_dx, xg.grad) < 1e-6
+
+
+# ------------------------------------------------------------------- loss --
+
+def _loss_inputs(seed=5):
+    g = torch.Generator().manual_seed(seed)
+    b, c, h, w = 2, 2, 32, 32
+    phm = torch.rand(b, c, h, w, generator=g).clamp(1e-4, 1 - 1e-4)
+    ghm = torch.rand(b, c, h, w, generator=g)
+    mask = (torch.rand(b, 1, h, w, generator=g) > 0.95).float()
+    poff = torch.randn(b, 2, h, w, generator=g)
+    goff = torch.rand(b, 2, h, w, generator=g)
+    psize = torch.randn(b, 2, h, w, generator=g) * 5
+    gsize = torch.rand(b, 2, h, w, generator=g) * 10
+    return phm, poff, psize, ghm, goff, gsize, mask
+
+
+def test_centernet_loss_fwd_bwd():
+    from real_time_helmet_detection_amd.ops import eager
+    ins = _loss_inputs()
+    phm, poff, psize, ghm, goff, gsize, mask = ins
+    want = eager.centernet_losses(phm, poff, psize, ghm, goff, gsize, mask,
+                                  2.0, 4.0)
+    gins = [t.cuda() for t in ins]
+    losses, sums = _C().centernet_loss_fwd(*gins, 2.0, 4.0)
+    for i in range(3):
+        assert abs(losses[i].item() - want[i].item()) < 1e-4 * max(
+            1.0, abs(want[i].item()))
+
+    # backward vs autograd of the eager op
+    phm_g = phm.clone().requires_grad_(True)
+    poff_g = poff.clone().requires_grad_(True)
+    psize_g = psize.clone().requires_grad_(True)
+    hm_l, off_l, size_l = eager.centernet_losses(
+        phm_g, poff_g, psize_g, ghm, goff, gsize, mask, 2.0, 4.0)
+    (hm_l * 1.0 + off_l * 2.0 + size_l * 0.5).backward()
+    gout = torch.tensor([1.0, 2.0, 0.5], device='cuda')
+    dphm, dpoff, dpsize = _C().centernet_loss_bwd(
+        *gins, sums, gout, 2.0, 4.0)
+    assert rel_err(dphm, phm_g.grad) < 1e-3
+    assert rel_err(dpoff, poff_g.grad) < 1e-5
+    assert rel_err(dpsize, psize_g.grad) < 1e-5
+
+
+# ----------------------------------------------------------------- decode --
+
+def test_decode_matches_eager():
+    from real_time_helmet_detection_amd.ops import eager
+    torch.manual_seed(6)
+    b, c, h, w = 3, 2, 64, 64
+    hm = torch.sigmoid(torch.randn(b, c, h, w) * 3)
+    off = torch.rand(b, 2, h, w)
+    wh = torch.rand(b, 2, h, w) * 20
+    want_boxes, want_cls, want_scores = eager.batched_decode(
+        hm, off, wh, 4, 50, 3, False)
+    boxes, clss, scores = _C().decode_fwd(hm.cuda(), off.cuda(), wh.cuda(),
+                                          4, 50, 3, False)
+    # order may differ on exact ties; compare sorted by score then x1
+    for i in range(b):
+        ws, wo = want_scores[i].sort(descending=True)
+        gs, go = scores[i].cpu().sort(descending=True)
+        torch.testing.assert_close(gs, ws, rtol=1e-5, atol=1e-6)
+        keep = ws > 0
+        torch.testing.assert_close(
+            boxes[i].cpu()[go][keep], want_boxes[i][wo][keep],
+            rtol=1e-4, atol=1e-4)
+        assert (clss[i].cpu()[go][keep] == want_cls[i][wo][keep]).all()
+
+
+def test_nms_matches_eager():
+    from real_time_helmet_detection_amd.ops import eager
+    torch.manual_seed(7)
+    xy = torch.rand(200, 2) * 100
+    wh = torch.rand(200, 2) * 30 + 1
+    boxes = torch.cat([xy, xy + wh], dim=1)
+    scores = torch.rand(200)
+    want = eager.nms(boxes, scores, 0.5)
+    got = _C().nms_fwd(boxes.cuda(), scores.cuda(), 0.5)
+    assert got.cpu().tolist() == want.tolist()
+
+
+# ------------------------------------------------------------------- conv --
+
+@pytest.mark.parametrize('cfg', [
+    # (Cin, Cout, k, stride, pad, H)
+    (32, 32, 3, 1, 1, 16),
+    (128, 128, 3, 1, 1, 16),
+    (64, 128, 1, 1, 0, 16),
+    (128, 6, 1, 1, 0, 16),
+    (128, 128, 2, 2, 0, 16),
+    (96, 160, 3, 1, 1, 12),   # non-pow2 channels, odd spatial
+])
+def test_conv_fwd_f32_exact(cfg):
+    cin, cout, k, stride, pad, h = cfg
+    torch.manual_seed(8)
+    x = torch.randn(2, cin, h, h)
+    w = torch.randn(cout, cin, k, k) * 0.1
+    bias = torch.randn(cout)
+    want = F.conv2d(x, w, bias, stride=stride, padding=pad)
+
+    wpk = _C().pack_weights(w.cuda(), False, False)
+    ones = torch.ones(cout, device='cuda')
+    got = _C().conv_fwd(to_gpu(x), wpk, ones, bias.cuda(), None, k, k,
+                        stride, pad, cout, 0)
+    assert rel_err(got, want) < 1e-5
+
+
+def test_conv_fwd_bf16_close():
+    torch.manual_seed(9)
+    x = torch.randn(2, 128, 16, 16)
+    w = torch.randn(128, 128, 3, 3) * 0.05
+    want = F.conv2d(x.to(torch.bfloat16).float(),
+                    w.to(torch.bfloat16).float(), None, padding=1)
+    wpk = _C().pack_weights(w.cuda(), False, True)
+    ones = torch.ones(128, device='cuda')
+    zeros = torch.zeros(128, device='cuda')
+    got = _C().conv_fwd(to_gpu(x, torch.bfloat16), wpk, ones, zeros, None,
+                        3, 3, 1, 1, 128, 0)
+    assert rel_err(got, want) < 0.03
+
+
+def test_conv_fwd_relu_epilogue():
+    torch.manual_seed(10)
+    x = torch.randn(1, 32, 8, 8)
+    w = torch.randn(32, 32, 3, 3) * 0.1
+    want = F.relu(F.conv2d(x, w, None, padding=1) * 2.0 + 0.5)
+    wpk = _C().pack_weights(w.cuda(), False, False)
+    sc = torch.full((32,), 2.0, device='cuda')
+    sh = torch.full((32,), 0.5, device='cuda')
+    got = _C().conv_fwd(to_gpu(x), wpk, sc, sh, None, 3, 3, 1, 1, 32, 1)
+    assert rel_err(got, want) < 1e-5
+
+
+def test_dgrad_via_swapped_pack():
+    # conv3x3 s1 p1: dX = conv(dY, rot180(W) transposed)
+    torch.manual_seed(11)
+    x = torch.randn(2, 64, 12, 12).requires_grad_(True)
+    w = torch.randn(32, 64, 3, 3) * 0.1
+    y = F.conv2d(x, w, None, padding=1)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    wpk_t = _C().pack_weights(w.cuda(), True, False)
+    ones = torch.ones(64, device='cuda')
+    zeros = torch.zeros(64, device='cuda')
+    got_dx = _C().conv_fwd(to_gpu(dy.detach()), wpk_t, ones, zeros, None,
+                           3, 3, 1, 1, 64, 0)
+    assert rel_err(got_dx, x.grad) < 1e-5
+
+
+@pytest.mark.parametrize('cfg', [
+    (64, 32, 3, 1, 1, 12),
+    (128, 128, 1, 1, 0, 16),
+    (3, 64, 7, 2, 3, 32),    # stem wgrad
+])
+def test_wgrad(cfg):
+    cin, cout, k, stride, pad, h = cfg
+    torch.manual_seed(12)
+    x = torch.randn(2, cin, h, h)
+    w = torch.randn(cout, cin, k, k, requires_grad=True)
+    y = F.conv2d(x, w, None, stride=stride, padding=pad)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    got = _C().wgrad(to_gpu(x), to_gpu(dy.detach()), k, k, stride, pad)
+    assert rel_err(got, w.grad) < 1e-4
+
+
+def test_stem_fwd():
+    torch.manual_seed(13)
+    x = torch.randn(2, 3, 64, 64)
+    w = torch.randn(64, 3, 7, 7) * 0.1
+    bias = torch.randn(64)
+    want = F.conv2d(x, w, bias, stride=2, padding=3)
+    ones = torch.ones(64, device='cuda')
+    got = _C().stem_fwd(to_gpu(x), w.cuda(), ones, bias.cuda(), 2, 3, 0)
+    assert rel_err(got, want) < 1e-5
+
+
+# -------------------------------------------------------------------- bn ---
+
+def test_bn_stats_and_fwd():
+    torch.manual_seed(14)
+    x = torch.randn(4, 32, 8, 8) * 3 + 1
+    rm = torch.zeros(32)
+    rv = torch.ones(32)
+    bn = torch.nn.BatchNorm2d(32)
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-0.5, 0.5)
+    want = F.relu(F.batch_norm(x, rm.clone(), rv.clone(), bn.weight, bn.bias,
+                               training=True, momentum=0.1, eps=1e-5))
+
+    rm_g, rv_g = rm.clone().cuda(), rv.clone().cuda()
+    mean, rstd = _C().bn_stats(to_gpu(x), rm_g, rv_g, 0.1, 1e-5)
+    got = _C().bn_act_fwd(to_gpu(x), mean, rstd,
+                          bn.weight.detach().cuda(),
+                          bn.bias.detach().cuda(), 1)
+    assert rel_err(got, want) < 1e-4
+    # running stats updated torch-style
+    rm_ref = rm.clone()
+    rv_ref = rv.clone()
+    F.batch_norm(x, rm_ref, rv_ref, bn.weight, bn.bias, training=True,
+                 momentum=0.1, eps=1e-5)
+    assert rel_err(rm_g, rm_ref) < 1e-4
+    assert rel_err(rv_g, rv_ref) < 1e-4
+
+
+def test_bn_act_bwd():
+    torch.manual_seed(15)
+    x = torch.randn(4, 32, 8, 8, requires_grad=True)
+    bn = torch.nn.BatchNorm2d(32)
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-0.5, 0.5)
+    y = F.relu(bn(x))
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    mean, rstd = _C().bn_stats(to_gpu(x.detach()), None, None, 0.1, 1e-5)
+    dx, dgamma, dbeta = _C().bn_act_bwd(to_gpu(dy), to_gpu(x.detach()),
+                                        mean, rstd,
+                                        bn.weight.detach().cuda(),
+                                        bn.bias.detach().cuda(), 1)
+    assert rel_err(dx, x.grad) < 1e-3
+    assert rel_err(dgamma, bn.weight.grad) < 1e-3
+    assert rel_err(dbeta, bn.bias.grad) < 1e-3
+
+
+def test_col_sum():
+    torch.manual_seed(16)
+    x = torch.randn(3, 33, 7, 9)  # odd sizes
+    want = x.sum(dim=(0, 2, 3))
+    got = _C().col_sum(to_gpu(x))
+    assert rel_err(got, want) < 1e-4

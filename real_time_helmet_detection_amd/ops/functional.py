@@ -28,13 +28,25 @@ def _maybe_bf16(x):
     return x
 
 
+def _named_act(y, act):
+    """Apply the fused-epilogue activation names on the eager path."""
+    if act == 'ReLU':
+        return F.relu(y)
+    if act == 'LReLU':
+        return F.leaky_relu(y, 0.01)
+    if act == 'Linear' or act is None:
+        return y
+    raise ValueError(f'non-fusible activation {act!r} must go through '
+                     'act_module')
+
+
 def conv_bn_act(x, conv, bn=None, act='Linear', act_module=None,
                 training=False):
     """conv -> (BN) -> activation, fused on the HIP path.
 
     conv: nn.Conv2d holding weight/bias; bn: nn.BatchNorm2d or None;
     act: activation name; act_module: the nn.Module for parametric /
-    non-fusible activations (PReLU etc.).
+    non-fusible activations (PReLU etc. — applied after a Linear epilogue).
     """
     if _hip(x):
         from . import hip
@@ -48,7 +60,7 @@ def conv_bn_act(x, conv, bn=None, act='Linear', act_module=None,
                          eps=bn.eps)
     if act_module is not None:
         return act_module(y)
-    return y
+    return _named_act(y, act)
 
 
 def add_act(a, b, act='Linear', act_module=None):
@@ -59,7 +71,7 @@ def add_act(a, b, act='Linear', act_module=None):
     y = a + b
     if act_module is not None:
         return act_module(y)
-    return y
+    return _named_act(y, act)
 
 
 def maxpool2x2(x):

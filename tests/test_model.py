@@ -92,3 +92,33 @@ def test_state_dict_key_layout():
     assert 'hourglass_lst.0.up1.conv1.convolution.weight' in keys
     assert 'neck_lst.0.layers.1.bn.running_mean' in keys
     assert 'head_lst.0.layer.convolution.bias' in keys
+
+
+def test_relu_actually_applied():
+    """Regression: the eager path once dropped the fused activation string —
+    a ReLU model's block outputs must be non-negative."""
+    from real_time_helmet_detection_amd.models import Convolution, Residual
+    torch.manual_seed(0)
+    m = Convolution(8, 8, 3, bias=False, bn=True, activation='ReLU').eval()
+    x = torch.randn(2, 8, 16, 16)
+    with torch.no_grad():
+        y = m(x)
+    assert (y >= 0).all()
+    assert (y > 0).any()
+    r = Residual(8, 8).eval()
+    with torch.no_grad():
+        yr = r(x)
+    assert (yr >= 0).all()
+
+
+def test_lrelu_slope():
+    from real_time_helmet_detection_amd.models import Convolution
+    import torch.nn.functional as F
+    torch.manual_seed(1)
+    m = Convolution(8, 8, 1, bias=True, bn=False, activation='LReLU').eval()
+    x = torch.randn(2, 8, 4, 4)
+    with torch.no_grad():
+        y = m(x)
+        want = F.leaky_relu(
+            F.conv2d(x, m.convolution.weight, m.convolution.bias), 0.01)
+    torch.testing.assert_close(y, want)

@@ -171,8 +171,15 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
       uint4 v = {0, 0, 0, 0};
       if (am[h] < g.M && iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
           c0 < g.Cin) {
-        v = *reinterpret_cast<const uint4*>(
-            x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0);
+        const bf16* src =
+            x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0;
+        if (c0 + 8 <= g.Cin) {
+          v = *reinterpret_cast<const uint4*>(src);
+        } else {  // chunk straddles Cin: masked element loads, zero pad
+          bf16 tmp[8] = {};
+          for (int e = 0; c0 + e < g.Cin; ++e) tmp[e] = src[e];
+          v = *reinterpret_cast<const uint4*>(tmp);
+        }
       }
       *reinterpret_cast<uint4*>(
           reinterpret_cast<char*>(A) +
@@ -302,7 +309,8 @@ void conv_fwd_f32_kernel(const float* __restrict__ x,
         const float* src =
             x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) v[e] = src[e];
+        for (int e = 0; e < 8; ++e)
+          v[e] = (c0 + e < g.Cin) ? src[e] : 0.f;
       }
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
@@ -394,8 +402,7 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
   g.Cinp = (int)cdiv(g.Cin, 32) * 32;
   g.Coutp = (int)cdiv(Cout, 128) * 128;
   g.M = g.B * g.Ho * g.Wo;
-  TORCH_CHECK(g.Cin % 32 == 0, "conv_fwd requires Cin % 32 == 0 (got ",
-              g.Cin, ") — the 3-channel stem uses stem_fwd");
+  TORCH_CHECK(g.Cin >= 1, "conv_fwd: bad Cin");
   TORCH_CHECK(wpk.size(0) == KH * KW && wpk.size(1) == g.Coutp &&
               wpk.size(2) == g.Cinp, "conv_fwd: packed weight shape");
 

@@ -150,12 +150,10 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
     stride = conv.stride[0]
     pad = conv.padding[0]
     cin = conv.weight.shape[1]
-    is_stem = cin < 32
-    if is_stem:
-        if (conv.weight.shape[0] != 64 or cin != 3 or kh != 7 or kw != 7):
-            raise NotImplementedError(
-                'HIP engine: small-Cin convs support only the 64x3x7x7 stem '
-                f'(got {tuple(conv.weight.shape)})')
+    # the 3-channel stem uses the direct kernel; everything else (any Cin)
+    # goes through the implicit-GEMM kernel
+    is_stem = (cin == 3 and conv.weight.shape[0] == 64 and kh == 7
+               and kw == 7)
     use_bn = bn is not None
     act_code = ACT_CODE.get(act)
     if act_code is None:

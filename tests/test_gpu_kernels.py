@@ -351,3 +351,17 @@ def test_col_sum():
     want = x.sum(dim=(0, 2, 3))
     got = _C().col_sum(to_gpu(x))
     assert rel_err(got, want) < 1e-4
+
+
+@pytest.mark.parametrize('cin', [6, 3, 20])
+def test_conv_fwd_small_cin(cin):
+    """Cin not a multiple of 32/8 (head dgrad, merge_prediction convs)."""
+    torch.manual_seed(17)
+    x = torch.randn(2, cin, 12, 12)
+    w = torch.randn(64, cin, 3, 3) * 0.1
+    want = F.conv2d(x, w, None, padding=1)
+    wpk = _C().pack_weights(w.cuda(), False, False)
+    ones = torch.ones(64, device='cuda')
+    zeros = torch.zeros(64, device='cuda')
+    got = _C().conv_fwd(to_gpu(x), wpk, ones, zeros, None, 3, 3, 1, 1, 64, 0)
+    assert rel_err(got, want) < 1e-5

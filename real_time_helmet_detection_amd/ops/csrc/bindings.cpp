@@ -46,6 +46,8 @@ torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
                        int64_t stride, int64_t pad, int64_t act);
 torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
                     int64_t KW, int64_t stride, int64_t pad);
+torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
+                              int64_t KW, int64_t stride, int64_t pad);
 
 std::vector<torch::Tensor> bn_stats(torch::Tensor x,
                                     c10::optional<torch::Tensor> running_mean,
@@ -78,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("stem_fwd", &rthd::stem_fwd);
   m.def("wgrad", &rthd::wgrad);
+  m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
   m.def("bn_stats", &rthd::bn_stats);
   m.def("bn_act_fwd", &rthd::bn_act_fwd);
   m.def("bn_act_bwd", &rthd::bn_act_bwd);

@@ -161,6 +161,204 @@ __global__ void bn_act_bwd_apply_kernel(
   }
 }
 
+
+// ---------------- vectorized (16 B/lane) fast paths, C % 8 == 0 ----------------
+
+// Row-interleaved octet assignment: thread t covers channel octet t%octs
+// of rows r0 + t/octs + k*streams — a wave reads contiguous full rows
+// (1 KB per wave-load, no over-fetch). Requires octs = C/8 a power of two
+// dividing 256 (the generic kernel covers everything else).
+template <typename T, bool WANT_SQ>
+__global__ void colsum8_kernel(const T* __restrict__ x,
+                               float* __restrict__ sum,
+                               float* __restrict__ sumsq,
+                               int64_t M, int C) {
+  const int octs = C >> 3;
+  const int streams = blockDim.x / octs;
+  const int oct = threadIdx.x % octs;
+  const int rs = threadIdx.x / octs;
+  const int c0 = oct * 8;
+  const int64_t rows_per_chunk = (M + gridDim.y - 1) / gridDim.y;
+  const int64_t r0 = blockIdx.y * rows_per_chunk;
+  const int64_t r1 = min(M, r0 + rows_per_chunk);
+  float acc[8] = {}, accsq[8] = {};
+  for (int64_t r = r0 + rs; r < r1; r += streams) {
+    T v[8];
+    *reinterpret_cast<uint4*>(v) =
+        *reinterpret_cast<const uint4*>(&x[r * C + c0]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float f = ldf(&v[e]);
+      acc[e] += f;
+      if (WANT_SQ) accsq[e] += f * f;
+    }
+  }
+  __shared__ float sh[256][8];
+  __shared__ float shq[WANT_SQ ? 256 : 1][WANT_SQ ? 8 : 1];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) sh[threadIdx.x][e] = acc[e];
+  if (WANT_SQ)
+#pragma unroll
+    for (int e = 0; e < 8; ++e) shq[threadIdx.x][e] = accsq[e];
+  __syncthreads();
+  for (int off = streams >> 1; off > 0; off >>= 1) {
+    if (rs < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        sh[threadIdx.x][e] += sh[threadIdx.x + off * octs][e];
+        if (WANT_SQ) shq[threadIdx.x][e] += shq[threadIdx.x + off * octs][e];
+      }
+    }
+    __syncthreads();
+  }
+  if (rs == 0) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      atomicAdd(&sum[c0 + e], sh[threadIdx.x][e]);
+      if (WANT_SQ) atomicAdd(&sumsq[c0 + e], shq[threadIdx.x][e]);
+    }
+  }
+}
+
+template <typename T>
+__global__ void bn_act_fwd8_kernel(const T* __restrict__ x,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   T* __restrict__ y, int64_t n8, int C,
+                                   int act) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n8; i += stride) {
+    const int c0 = (int)((i * 8) % C);
+    T v[8], o[8];
+    *reinterpret_cast<uint4*>(v) =
+        *reinterpret_cast<const uint4*>(&x[i * 8]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int c = c0 + e;
+      const float xh = (ldf(&v[e]) - mean[c]) * rstd[c];
+      stf(&o[e], apply_act(xh * gamma[c] + beta[c], act));
+    }
+    *reinterpret_cast<uint4*>(&y[i * 8]) = *reinterpret_cast<uint4*>(o);
+  }
+}
+
+template <typename T>
+__global__ void bn_act_bwd_reduce8_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ s1, float* __restrict__ s2,
+    int64_t M, int C, int act) {
+  const int octs = C >> 3;
+  const int streams = blockDim.x / octs;
+  const int oct = threadIdx.x % octs;
+  const int rstream = threadIdx.x / octs;
+  const int c0 = oct * 8;
+  const int64_t rows_per_chunk = (M + gridDim.y - 1) / gridDim.y;
+  const int64_t r0 = blockIdx.y * rows_per_chunk;
+  const int64_t r1 = min(M, r0 + rows_per_chunk);
+  float mu[8], rs[8], gm[8], bt[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    mu[e] = mean[c0 + e];
+    rs[e] = rstd[c0 + e];
+    gm[e] = gamma[c0 + e];
+    bt[e] = beta[c0 + e];
+  }
+  float a1[8] = {}, a2[8] = {};
+  for (int64_t r = r0 + rstream; r < r1; r += streams) {
+    T vx[8], vdy[8];
+    *reinterpret_cast<uint4*>(vx) =
+        *reinterpret_cast<const uint4*>(&x[r * C + c0]);
+    *reinterpret_cast<uint4*>(vdy) =
+        *reinterpret_cast<const uint4*>(&dy[r * C + c0]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float xh = (ldf(&vx[e]) - mu[e]) * rs[e];
+      const float pre = xh * gm[e] + bt[e];
+      const float dpre = ldf(&vdy[e]) *
+          (act == ACT_RELU ? (pre > 0.f ? 1.f : 0.f)
+                           : (act == ACT_LRELU ? (pre > 0.f ? 1.f : 0.01f)
+                                               : 1.f));
+      a1[e] += dpre;
+      a2[e] += dpre * xh;
+    }
+  }
+  __shared__ float sh1[256][8], sh2[256][8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    sh1[threadIdx.x][e] = a1[e];
+    sh2[threadIdx.x][e] = a2[e];
+  }
+  __syncthreads();
+  for (int off = streams >> 1; off > 0; off >>= 1) {
+    if (rstream < off) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        sh1[threadIdx.x][e] += sh1[threadIdx.x + off * octs][e];
+        sh2[threadIdx.x][e] += sh2[threadIdx.x + off * octs][e];
+      }
+    }
+    __syncthreads();
+  }
+  if (rstream == 0) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      atomicAdd(&s1[c0 + e], sh1[threadIdx.x][e]);
+      atomicAdd(&s2[c0 + e], sh2[threadIdx.x][e]);
+    }
+  }
+}
+
+template <typename T>
+__global__ void bn_act_bwd_apply8_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ s1, const float* __restrict__ s2,
+    T* __restrict__ dx, int64_t n8, int C, float Mf, int act) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float invM = 1.f / Mf;
+  for (; i < n8; i += stride) {
+    const int c0 = (int)((i * 8) % C);
+    T vx[8], vdy[8], o[8];
+    *reinterpret_cast<uint4*>(vx) =
+        *reinterpret_cast<const uint4*>(&x[i * 8]);
+    *reinterpret_cast<uint4*>(vdy) =
+        *reinterpret_cast<const uint4*>(&dy[i * 8]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int c = c0 + e;
+      const float xh = (ldf(&vx[e]) - mean[c]) * rstd[c];
+      const float pre = xh * gamma[c] + beta[c];
+      const float dpre = ldf(&vdy[e]) *
+          (act == ACT_RELU ? (pre > 0.f ? 1.f : 0.f)
+                           : (act == ACT_LRELU ? (pre > 0.f ? 1.f : 0.01f)
+                                               : 1.f));
+      stf(&o[e], gamma[c] * rstd[c] *
+          (dpre - s1[c] * invM - xh * s2[c] * invM));
+    }
+    *reinterpret_cast<uint4*>(&dx[i * 8]) = *reinterpret_cast<uint4*>(o);
+  }
+}
+
+static bool fast8_ok(int64_t M, int C) {
+  const int octs = C / 8;
+  return C % 8 == 0 && octs > 0 && octs <= 256 &&
+         (octs & (octs - 1)) == 0 && 256 % octs == 0;
+}
+
+static dim3 red_grid8(int64_t M, int C) {
+  // one block covers all channels; chunk rows to ~2048 blocks
+  int chunks = (int)std::min<int64_t>(
+      std::max<int64_t>(M * C / (256 * 8), 1), 2048);
+  return dim3(1, chunks);
+}
+
 // --------------------------------- wrappers ---------------------------------
 
 static dim3 red_grid(int64_t M, int C) {
@@ -192,9 +390,14 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   auto rstd = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
   DT(xc, {
-    hipLaunchKernelGGL((colsum_kernel<scalar_t, true>), red_grid(M, C),
-        dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-        sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C);
+    if (fast8_ok(M, C))
+      hipLaunchKernelGGL((colsum8_kernel<scalar_t, true>), red_grid8(M, C),
+          dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C);
+    else
+      hipLaunchKernelGGL((colsum_kernel<scalar_t, true>), red_grid(M, C),
+          dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C);
   });
   float* rm = nullptr;
   float* rv = nullptr;
@@ -222,12 +425,20 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
   auto bt = beta.to(at::kFloat).contiguous();
   auto s = at::cuda::getCurrentCUDAStream();
   DT(xc, {
-    hipLaunchKernelGGL((bn_act_fwd_kernel<scalar_t>),
-        dim3(ew_grid(n, 256)), dim3(256), 0, s,
-        reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        reinterpret_cast<scalar_t*>(y.data_ptr()), n, C, (int)act);
+    if (C % 8 == 0)
+      hipLaunchKernelGGL((bn_act_fwd8_kernel<scalar_t>),
+          dim3(ew_grid(n / 8, 256)), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          reinterpret_cast<scalar_t*>(y.data_ptr()), n / 8, C, (int)act);
+    else
+      hipLaunchKernelGGL((bn_act_fwd_kernel<scalar_t>),
+          dim3(ew_grid(n, 256)), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          reinterpret_cast<scalar_t*>(y.data_ptr()), n, C, (int)act);
   });
   HIP_CHECK_LAST();
   return y;
@@ -249,22 +460,41 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   auto bt = beta.to(at::kFloat).contiguous();
   auto s = at::cuda::getCurrentCUDAStream();
   DT(xc, {
-    hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<scalar_t>),
-        red_grid(M, C), dim3(256), 0, s,
-        reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-        reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        s1.data_ptr<float>(), s2.data_ptr<float>(), M, C, (int)act);
-    hipLaunchKernelGGL((bn_act_bwd_apply_kernel<scalar_t>),
-        dim3(ew_grid(n, 256)), dim3(256), 0, s,
-        reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-        reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        s1.data_ptr<float>(), s2.data_ptr<float>(),
-        reinterpret_cast<scalar_t*>(dx.data_ptr()), n, C, (float)M,
-        (int)act);
+    if (fast8_ok(M, C)) {
+      hipLaunchKernelGGL((bn_act_bwd_reduce8_kernel<scalar_t>),
+          red_grid8(M, C), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          s1.data_ptr<float>(), s2.data_ptr<float>(), M, C, (int)act);
+      hipLaunchKernelGGL((bn_act_bwd_apply8_kernel<scalar_t>),
+          dim3(ew_grid(n / 8, 256)), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          s1.data_ptr<float>(), s2.data_ptr<float>(),
+          reinterpret_cast<scalar_t*>(dx.data_ptr()), n / 8, C, (float)M,
+          (int)act);
+    } else {
+      hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<scalar_t>),
+          red_grid(M, C), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          s1.data_ptr<float>(), s2.data_ptr<float>(), M, C, (int)act);
+      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<scalar_t>),
+          dim3(ew_grid(n, 256)), dim3(256), 0, s,
+          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
+          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          s1.data_ptr<float>(), s2.data_ptr<float>(),
+          reinterpret_cast<scalar_t*>(dx.data_ptr()), n, C, (float)M,
+          (int)act);
+    }
   });
   HIP_CHECK_LAST();
   // dgamma = s2, dbeta = s1
@@ -279,9 +509,14 @@ torch::Tensor col_sum(torch::Tensor x) {
   auto sum = torch::zeros({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
   DT(xc, {
-    hipLaunchKernelGGL((colsum_kernel<scalar_t, false>), red_grid(M, C),
-        dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-        sum.data_ptr<float>(), nullptr, M, C);
+    if (fast8_ok(M, C))
+      hipLaunchKernelGGL((colsum8_kernel<scalar_t, false>), red_grid8(M, C),
+          dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          sum.data_ptr<float>(), nullptr, M, C);
+    else
+      hipLaunchKernelGGL((colsum_kernel<scalar_t, false>), red_grid(M, C),
+          dim3(256), 0, s, reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+          sum.data_ptr<float>(), nullptr, M, C);
   });
   HIP_CHECK_LAST();
   return sum;

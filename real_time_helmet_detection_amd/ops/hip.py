@@ -138,7 +138,10 @@ class _ConvBNActFn(torch.autograd.Function):
         # wgrad
         dw = None
         if ctx.needs_input_grad[1]:
-            dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
+            if xc.dtype == torch.bfloat16:
+                dw = C.wgrad_bf16_fast(xc, dpre, kh, kw, stride, pad)
+            else:
+                dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
 
         return (dx, dw, dbias, dgamma, dbeta, None, None,
                 None, None, None, None, None, None, None, None, None, None)

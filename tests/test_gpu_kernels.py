@@ -365,3 +365,23 @@ def test_conv_fwd_small_cin(cin):
     zeros = torch.zeros(64, device='cuda')
     got = _C().conv_fwd(to_gpu(x), wpk, ones, zeros, None, 3, 3, 1, 1, 64, 0)
     assert rel_err(got, want) < 1e-5
+
+
+@pytest.mark.parametrize('cfg', [
+    (64, 32, 3, 1, 1, 12),
+    (128, 128, 3, 1, 1, 16),
+    (128, 128, 1, 1, 0, 16),
+    (3, 64, 7, 2, 3, 32),
+    (128, 6, 1, 1, 0, 16),
+])
+def test_wgrad_bf16_fast(cfg):
+    cin, cout, k, stride, pad, h = cfg
+    torch.manual_seed(18)
+    x = (torch.randn(2, cin, h, h)).to(torch.bfloat16).float()
+    w = torch.randn(cout, cin, k, k, requires_grad=True)
+    y = F.conv2d(x, w, None, stride=stride, padding=pad)
+    dy = torch.randn_like(y).to(torch.bfloat16).float()
+    y.backward(dy)
+    got = _C().wgrad_bf16_fast(to_gpu(x, torch.bfloat16),
+                               to_gpu(dy, torch.bfloat16), k, k, stride, pad)
+    assert rel_err(got, w.grad) < 0.03

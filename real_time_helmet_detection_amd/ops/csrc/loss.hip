@@ -153,7 +153,10 @@ std::vector<torch::Tensor> centernet_loss_fwd(
   auto losses = torch::empty({3}, phm_.options());
   auto s = at::cuda::getCurrentCUDAStream();
   const int64_t n = (int64_t)B * C * HW;
-  hipLaunchKernelGGL(centernet_loss_sums_kernel, dim3(ew_grid(n, 256)),
+  // cap blocks: every block ends with 5 same-address atomics (guide G12)
+  int nblk = ew_grid(n, 256);
+  if (nblk > 512) nblk = 512;
+  hipLaunchKernelGGL(centernet_loss_sums_kernel, dim3(nblk),
       dim3(256), 0, s,
       phm_.data_ptr<float>(), ghm_.data_ptr<float>(),
       poff_.data_ptr<float>(), goff_.data_ptr<float>(),

@@ -133,21 +133,37 @@ __global__ void colsum8_part_kernel(const bf16* __restrict__ x,
   }
 }
 
-// out[c] = sum_k p[k][c]  (and out2/p2 when given)
+// out[c] = sum_k p[k][c]  (and out2/p2 when given).
+// block = 64 channels x 4 k-substreams (a K=256 serial loop in one tiny
+// block was 44 us — latency-bound).
 __global__ void reduce_partials_kernel(const float* __restrict__ p1,
                                        const float* __restrict__ p2,
                                        float* __restrict__ o1,
                                        float* __restrict__ o2,
                                        int K, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int sub = threadIdx.x >> 6;
   float a = 0.f, b = 0.f;
-  for (int k = 0; k < K; ++k) {
-    a += p1[(int64_t)k * C + c];
-    if (p2) b += p2[(int64_t)k * C + c];
+  if (c < C) {
+    for (int k = sub; k < K; k += 4) {
+      a += p1[(int64_t)k * C + c];
+      if (p2) b += p2[(int64_t)k * C + c];
+    }
   }
-  o1[c] = a;
-  if (p2) o2[c] = b;
+  __shared__ float sha[256], shb[256];
+  sha[threadIdx.x] = a;
+  shb[threadIdx.x] = b;
+  __syncthreads();
+  if (sub == 0 && c < C) {
+    a = sha[threadIdx.x] + sha[threadIdx.x + 64] + sha[threadIdx.x + 128] +
+        sha[threadIdx.x + 192];
+    o1[c] = a;
+    if (p2) {
+      b = shb[threadIdx.x] + shb[threadIdx.x + 64] + shb[threadIdx.x + 128] +
+          shb[threadIdx.x + 192];
+      o2[c] = b;
+    }
+  }
 }
 
 // finalize mean/rstd (+ running-stat update, torch semantics)
@@ -433,7 +449,7 @@ static void run_colsum(const torch::Tensor& xc, torch::Tensor& sum,
             p1.data_ptr<float>(), p2p, M, C);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 256)), dim3(256),
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64)), dim3(256),
       0, s, p1.data_ptr<float>(), p2p, sum.data_ptr<float>(),
       sumsq ? sumsq->data_ptr<float>() : nullptr, chunks, C);
 }
@@ -537,7 +553,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
           p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 256)), dim3(256),
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64)), dim3(256),
       0, s, p1.data_ptr<float>(), p2.data_ptr<float>(),
       s1.data_ptr<float>(), s2.data_ptr<float>(), chunks, C);
 

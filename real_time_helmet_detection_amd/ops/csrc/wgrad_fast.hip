@@ -1,17 +1,18 @@
-// Fast bf16 wgrad: register-transposed staging.
+// Fast bf16 wgrad: register-transposed staging, v3.
 //
-// The generic wgrad kernel (wgrad.hip) stages the transposed [ch][px] LDS
-// images with 16 scalar ds_write_b16 per thread per 32-px step — measured
-// 40% of the training step. This kernel loads 8px x 4ci per thread,
-// transposes IN REGISTERS (the compiler lowers the short shuffles to
-// v_perm/pack ops on the 32-wide VALU) and writes FOUR ds_write_b128 per
-// tile — 8x fewer LDS write instructions — while staging a 128-px K-block
-// (4 MFMA k-steps per barrier instead of 1).
-//
-// Layout: X image [64 ci][128 px] bf16 rows of 256 B with an XOR slot
-// swizzle (slot' = slot ^ (ci & 7), 16 slots/row) so the 16-lane fragment
-// read groups spread banks; same for dY. Fragments then read with
-// ds_read_b128 exactly like conv.hip.
+// dW_t[ci][co] = sum_m X_t[m][ci] * dY[m][co] as an MFMA GEMM with the
+// pixel dim as K. Design points (each measured against the previous
+// version):
+// - Transposed [ch][px] LDS images built by an 8x8 IN-REGISTER transpose
+//   and ds_write_b128 (the v1 16x ds_write_b16 scatter was 40% of the
+//   whole training step).
+// - Thread halves split the work: threads 0..127 stage the X tile while
+//   128..255 stage dY — each thread 8 px x 8 ch with 16-B loads.
+// - Pixel->(b,oy,ox) decomposition is INCREMENTAL int32 (one div pair per
+//   block launch; +128 carry walk per round) — the per-round int64 div/mod
+//   chains of v2 serialized the loop.
+// - 128-px K block = 4 MFMA k-steps per barrier pair.
+// fp32 falls back to the generic kernel in wgrad.hip.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
@@ -21,9 +22,9 @@ namespace rthd {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-// [64 rows][128 px] bf16: row stride 256 B = 16 slots of 16 B.
-DEV_INLINE int wg_off(int row, int px8 /*0..15*/) {
-  // 16 slots x 16 rows XOR -> conflict-free fragment reads
+// [64 rows][128 px] bf16: row stride 256 B = 16 slots of 16 B;
+// slot' = slot ^ (row & 15) -> conflict-free 16-lane fragment reads.
+DEV_INLINE int wg_off(int row, int px8) {
   return row * 256 + ((px8 ^ (row & 15)) << 4);
 }
 
@@ -59,45 +60,54 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   const int px_start = chunk * g.chunk_len;
   const int px_end = min(px_start + g.chunk_len, g.M);
 
-  // staging assignment: thread -> (ci_oct = tid%16 -> 4 ci, px_blk =
-  // tid/16 -> 8 px); covers 64 ci x 128 px per tile.
-  const int s_ci = (tid & 15) * 4;
-  const int s_px = (tid >> 4) * 8;
+  // staging role: half 0 -> X, half 1 -> dY; within a half:
+  // ci octet = tid & 7 (8 ci), px block = (tid >> 3) & 15 (8 px)
+  const bool stage_x = tid < 128;
+  const int s_ci = (tid & 7) * 8;
+  const int s_px = ((tid >> 3) & 15) * 8;
+
+  // incremental (b, oy, ox) for this thread's px base (X half only)
+  const int HoWo = g.Ho * g.Wo;
+  int mb = px_start + s_px;
+  int bb = mb / HoWo;
+  int rr = mb - bb * HoWo;
+  int oy = rr / g.Wo;
+  int ox = rr - oy * g.Wo;
 
   for (int p0 = px_start; p0 < px_end; p0 += 128) {
     __syncthreads();
-
-    // ---- stage X_t ----
-    {
-      ushort r[8][4];
+    ushort r[8][8];
+    if (stage_x) {
+      int bj = bb, oyj = oy, oxj = ox;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int m = p0 + s_px + j;
-        uint2 v = {0, 0};
+        uint4 v = {0, 0, 0, 0};
         if (m < px_end) {
-          const int b = m / (g.Ho * g.Wo);
-          const int rr = m % (g.Ho * g.Wo);
-          const int iy = (rr / g.Wo) * g.stride + dyt;
-          const int ix = (rr % g.Wo) * g.stride + dxt;
-          const int cbase = ci0 + s_ci;
-          if (iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
-              cbase + 4 <= g.Cin) {
-            v = *reinterpret_cast<const uint2*>(
-                x + (((int64_t)b * g.H + iy) * g.W + ix) * g.Cin + cbase);
-          } else if (iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
-                     cbase < g.Cin) {
-            const bf16* src =
-                x + (((int64_t)b * g.H + iy) * g.W + ix) * g.Cin + cbase;
-            ushort tmp[4] = {};
-            for (int e = 0; cbase + e < g.Cin; ++e)
-              tmp[e] = reinterpret_cast<const ushort*>(src)[e];
-            v = *reinterpret_cast<const uint2*>(tmp);
+          const int iy = oyj * g.stride + dyt;
+          const int ix = oxj * g.stride + dxt;
+          if (iy >= 0 && iy < g.H && ix >= 0 && ix < g.W) {
+            const int cbase = ci0 + s_ci;
+            if (cbase + 8 <= g.Cin) {
+              v = *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
+            } else if (cbase < g.Cin) {
+              const ushort* src = reinterpret_cast<const ushort*>(
+                  x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
+              ushort tmp[8] = {};
+              for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
+              v = *reinterpret_cast<const uint4*>(tmp);
+            }
           }
         }
-        *reinterpret_cast<uint2*>(r[j]) = v;
+        *reinterpret_cast<uint4*>(r[j]) = v;
+        if (++oxj >= g.Wo) {
+          oxj = 0;
+          if (++oyj >= g.Ho) { oyj = 0; ++bj; }
+        }
       }
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
+      for (int e = 0; e < 8; ++e) {
         ushort o[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) o[j] = r[j][e];
@@ -105,32 +115,29 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
             reinterpret_cast<char*>(Xl) +
             wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
       }
-    }
-
-    // ---- stage dY ----
-    {
-      ushort r[8][4];
+    } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int m = p0 + s_px + j;
-        uint2 v = {0, 0};
+        uint4 v = {0, 0, 0, 0};
         if (m < px_end) {
           const int cbase = co0 + s_ci;
-          if (cbase + 4 <= g.Cout) {
-            v = *reinterpret_cast<const uint2*>(
+          if (cbase + 8 <= g.Cout) {
+            v = *reinterpret_cast<const uint4*>(
                 dy + (int64_t)m * g.Cout + cbase);
           } else if (cbase < g.Cout) {
-            const bf16* src = dy + (int64_t)m * g.Cout + cbase;
-            ushort tmp[4] = {};
-            for (int e = 0; cbase + e < g.Cout; ++e)
-              tmp[e] = reinterpret_cast<const ushort*>(src)[e];
-            v = *reinterpret_cast<const uint2*>(tmp);
+            const ushort* src =
+                reinterpret_cast<const ushort*>(dy + (int64_t)m * g.Cout +
+                                                cbase);
+            ushort tmp[8] = {};
+            for (int e = 0; cbase + e < g.Cout; ++e) tmp[e] = src[e];
+            v = *reinterpret_cast<const uint4*>(tmp);
           }
         }
-        *reinterpret_cast<uint2*>(r[j]) = v;
+        *reinterpret_cast<uint4*>(r[j]) = v;
       }
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
+      for (int e = 0; e < 8; ++e) {
         ushort o[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) o[j] = r[j][e];
@@ -139,12 +146,19 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
             wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
       }
     }
+    // advance the X half's incremental decomposition by 128 px
+    if (stage_x) {
+      ox += 128;
+      while (ox >= g.Wo) {
+        ox -= g.Wo;
+        if (++oy >= g.Ho) { oy = 0; ++bb; }
+      }
+    }
     __syncthreads();
 
-    // ---- 4 MFMA k-steps over the 128-px block ----
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
-      const int k8 = (lane >> 4) + ks * 4;  // 16-B slot index (8 px)
+      const int k8 = (lane >> 4) + ks * 4;
       bf16x8 xa[2], yb[2];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
@@ -168,15 +182,15 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int ci = ci0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r;
+    for (int r2 = 0; r2 < 4; ++r2) {
+      const int ci = ci0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r2;
       if (ci >= g.Cin) continue;
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
         const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
         atomicAdd(&dw[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx],
-                  acc[mi][ni][r]);
+                  acc[mi][ni][r2]);
       }
     }
   }

@@ -44,6 +44,8 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
 torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
                        torch::Tensor scale, torch::Tensor shift,
                        int64_t stride, int64_t pad, int64_t act);
+torch::Tensor stem_wgrad(torch::Tensor x, torch::Tensor dy, int64_t stride,
+                         int64_t pad);
 torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
                     int64_t KW, int64_t stride, int64_t pad);
 torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
@@ -79,6 +81,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_weights", &rthd::pack_weights);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("stem_fwd", &rthd::stem_fwd);
+  m.def("stem_wgrad", &rthd::stem_wgrad);
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
   m.def("bn_stats", &rthd::bn_stats);

@@ -127,8 +127,8 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
   const int wr = wid >> 1, wc = wid & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* ldsA = reinterpret_cast<bf16*>(smem);             // 2 x 8 KB
-  bf16* ldsB = reinterpret_cast<bf16*>(smem + 16384);     // 2 x 8 KB
+  bf16* ldsA = reinterpret_cast<bf16*>(smem);             // 8 KB
+  bf16* ldsB = reinterpret_cast<bf16*>(smem + 8192);      // 8 KB
 
   f32x4 acc[4][4] = {};
 
@@ -153,16 +153,16 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
   const int nsteps = g.KH * g.KW * (g.Cinp / 32);
   const int kc_per_tap = g.Cinp / 32;
 
-  for (int step = 0; step < nsteps; ++step) {
+  // register-staged software pipeline (guide G15/T14): loads for step k+1
+  // are ISSUED during step k's MFMA and only CONSUMED at the next ds_write,
+  // hiding the HBM latency that a stage->sync->MFMA loop exposes every
+  // K-step. Single LDS buffer (the write of k+1 happens after the second
+  // barrier, when step k's fragment reads are done).
+  auto load_step = [&](int step, uint4 (&va)[2], uint4 (&vb)[2]) {
     const int t = step / kc_per_tap;
     const int kb = step % kc_per_tap;
     const int dy = t / g.KW - g.pad;
     const int dx = t % g.KW - g.pad;
-    const int buf = step & 1;
-    bf16* A = ldsA + buf * 4096;
-    bf16* B = ldsB + buf * 4096;
-
-    // ---- stage A (2 rows per thread) ----
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int iy = ay[h] * g.stride + dy;
@@ -181,24 +181,34 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
           v = *reinterpret_cast<const uint4*>(tmp);
         }
       }
-      *reinterpret_cast<uint4*>(
-          reinterpret_cast<char*>(A) +
-          lds_off_bf16(st_row + 64 * h, st_k8)) = v;
+      va[h] = v;
+      const int64_t src_off =
+          ((int64_t)t * g.Coutp + nblk * 128 + st_row + 64 * h) * g.Cinp +
+          kb * 32 + st_k8 * 8;
+      vb[h] = *reinterpret_cast<const uint4*>(wpk + src_off);
     }
+  };
 
-    // ---- stage B (2 rows per thread) ----
+  bf16* A = ldsA;
+  bf16* B = ldsB;
+  uint4 sa[2], sb[2], na[2], nb[2];
+  load_step(0, sa, sb);
+
+  for (int step = 0; step < nsteps; ++step) {
+    // ---- write current tile (regs -> LDS) ----
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      const int row = st_row + 64 * h;  // cout within tile
-      const int64_t src_off =
-          ((int64_t)t * g.Coutp + nblk * 128 + row) * g.Cinp + kb * 32 +
-          st_k8 * 8;
-      const uint4 v = *reinterpret_cast<const uint4*>(wpk + src_off);
       *reinterpret_cast<uint4*>(
-          reinterpret_cast<char*>(B) + lds_off_bf16(row, st_k8)) = v;
+          reinterpret_cast<char*>(A) +
+          lds_off_bf16(st_row + 64 * h, st_k8)) = sa[h];
+      *reinterpret_cast<uint4*>(
+          reinterpret_cast<char*>(B) + lds_off_bf16(st_row + 64 * h,
+                                                    st_k8)) = sb[h];
     }
-
     __syncthreads();
+
+    // ---- issue next tile's loads (consumed at the next ds_write) ----
+    if (step + 1 < nsteps) load_step(step + 1, na, nb);
 
     // ---- fragments + MFMA ----
     const int arow_base = wr * 64 + (lane & 15);
@@ -220,6 +230,11 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
 
     __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      sa[h] = na[h];
+      sb[h] = nb[h];
+    }
   }
 
   // ---- epilogue ----
@@ -423,7 +438,7 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
   auto s = at::cuda::getCurrentCUDAStream();
 
   if (bf16_mode) {
-    const size_t lds = 32768;
+    const size_t lds = 16384;
     if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
     auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
     auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());

@@ -138,7 +138,9 @@ class _ConvBNActFn(torch.autograd.Function):
         # wgrad
         dw = None
         if ctx.needs_input_grad[1]:
-            if xc.dtype == torch.bfloat16:
+            if is_stem:
+                dw = C.stem_wgrad(xc, dpre, stride, pad)
+            elif xc.dtype == torch.bfloat16:
                 dw = C.wgrad_bf16_fast(xc, dpre, kh, kw, stride, pad)
             else:
                 dw = C.wgrad(xc, dpre, kh, kw, stride, pad)

@@ -385,3 +385,20 @@ def test_wgrad_bf16_fast(cfg):
     got = _C().wgrad_bf16_fast(to_gpu(x, torch.bfloat16),
                                to_gpu(dy, torch.bfloat16), k, k, stride, pad)
     assert rel_err(got, w.grad) < 0.03
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_stem_wgrad(dtype):
+    torch.manual_seed(19)
+    x = torch.randn(2, 3, 64, 64)
+    if dtype == torch.bfloat16:
+        x = x.to(dtype).float()
+    w = torch.randn(64, 3, 7, 7, requires_grad=True)
+    y = F.conv2d(x, w, None, stride=2, padding=3)
+    dy = torch.randn_like(y)
+    if dtype == torch.bfloat16:
+        dy = dy.to(dtype).float()
+    y.backward(dy)
+    got = _C().stem_wgrad(to_gpu(x, dtype), to_gpu(dy, dtype), 2, 3)
+    tol = 0.02 if dtype == torch.bfloat16 else 1e-4
+    assert rel_err(got, w.grad) < tol

@@ -157,58 +157,77 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
   // are ISSUED during step k's MFMA and only CONSUMED at the next ds_write,
   // hiding the HBM latency that a stage->sync->MFMA loop exposes every
   // K-step. Single LDS buffer (the write of k+1 happens after the second
-  // barrier, when step k's fragment reads are done).
-  auto load_step = [&](int step, uint4 (&va)[2], uint4 (&vb)[2]) {
-    const int t = step / kc_per_tap;
-    const int kb = step % kc_per_tap;
-    const int dy = t / g.KW - g.pad;
-    const int dx = t % g.KW - g.pad;
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      const int iy = ay[h] * g.stride + dy;
-      const int ix = ax[h] * g.stride + dx;
-      const int c0 = kb * 32 + st_k8 * 8;
-      uint4 v = {0, 0, 0, 0};
-      if (am[h] < g.M && iy >= 0 && iy < g.H && ix >= 0 && ix < g.W &&
-          c0 < g.Cin) {
-        const bf16* src =
-            x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0;
-        if (c0 + 8 <= g.Cin) {
-          v = *reinterpret_cast<const uint4*>(src);
-        } else {  // chunk straddles Cin: masked element loads, zero pad
-          bf16 tmp[8] = {};
-          for (int e = 0; c0 + e < g.Cin; ++e) tmp[e] = src[e];
-          v = *reinterpret_cast<const uint4*>(tmp);
-        }
-      }
-      va[h] = v;
-      const int64_t src_off =
-          ((int64_t)t * g.Coutp + nblk * 128 + st_row + 64 * h) * g.Cinp +
-          kb * 32 + st_k8 * 8;
-      vb[h] = *reinterpret_cast<const uint4*>(wpk + src_off);
-    }
-  };
+  // barrier, when step k's fragment reads are done). Scalars, not arrays:
+  // by-reference arrays in a lambda spilled to scratch (guide rule 20).
+#define RTHD_LOAD_STEP(STEP, A0, A1, B0, B1)                                 \
+  do {                                                                       \
+    const int t_ = (STEP) / kc_per_tap;                                      \
+    const int kb_ = (STEP) % kc_per_tap;                                     \
+    const int dy_ = t_ / g.KW - g.pad;                                       \
+    const int dx_ = t_ % g.KW - g.pad;                                       \
+    const int c0_ = kb_ * 32 + st_k8 * 8;                                    \
+    uint4 v0_ = {0, 0, 0, 0}, v1_ = {0, 0, 0, 0};                            \
+    {                                                                        \
+      const int iy_ = ay[0] * g.stride + dy_;                                \
+      const int ix_ = ax[0] * g.stride + dx_;                                \
+      if (am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
+          c0_ < g.Cin) {                                                     \
+        const bf16* src_ =                                                   \
+            x + (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
+        if (c0_ + 8 <= g.Cin) {                                              \
+          v0_ = *reinterpret_cast<const uint4*>(src_);                       \
+        } else {                                                             \
+          bf16 tmp_[8] = {};                                                 \
+          for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];      \
+          v0_ = *reinterpret_cast<const uint4*>(tmp_);                       \
+        }                                                                    \
+      }                                                                      \
+    }                                                                        \
+    {                                                                        \
+      const int iy_ = ay[1] * g.stride + dy_;                                \
+      const int ix_ = ax[1] * g.stride + dx_;                                \
+      if (am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
+          c0_ < g.Cin) {                                                     \
+        const bf16* src_ =                                                   \
+            x + (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
+        if (c0_ + 8 <= g.Cin) {                                              \
+          v1_ = *reinterpret_cast<const uint4*>(src_);                       \
+        } else {                                                             \
+          bf16 tmp_[8] = {};                                                 \
+          for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];      \
+          v1_ = *reinterpret_cast<const uint4*>(tmp_);                       \
+        }                                                                    \
+      }                                                                      \
+    }                                                                        \
+    (A0) = v0_;                                                              \
+    (A1) = v1_;                                                              \
+    (B0) = *reinterpret_cast<const uint4*>(                                  \
+        wpk + ((int64_t)t_ * g.Coutp + nblk * 128 + st_row) * g.Cinp +       \
+        kb_ * 32 + st_k8 * 8);                                               \
+    (B1) = *reinterpret_cast<const uint4*>(                                  \
+        wpk + ((int64_t)t_ * g.Coutp + nblk * 128 + st_row + 64) * g.Cinp +  \
+        kb_ * 32 + st_k8 * 8);                                               \
+  } while (0)
 
   bf16* A = ldsA;
   bf16* B = ldsB;
-  uint4 sa[2], sb[2], na[2], nb[2];
-  load_step(0, sa, sb);
+  uint4 sa0, sa1, sb0, sb1, na0, na1, nb0, nb1;
+  RTHD_LOAD_STEP(0, sa0, sa1, sb0, sb1);
 
   for (int step = 0; step < nsteps; ++step) {
     // ---- write current tile (regs -> LDS) ----
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      *reinterpret_cast<uint4*>(
-          reinterpret_cast<char*>(A) +
-          lds_off_bf16(st_row + 64 * h, st_k8)) = sa[h];
-      *reinterpret_cast<uint4*>(
-          reinterpret_cast<char*>(B) + lds_off_bf16(st_row + 64 * h,
-                                                    st_k8)) = sb[h];
-    }
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(A) + lds_off_bf16(st_row, st_k8)) = sa0;
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(A) + lds_off_bf16(st_row + 64, st_k8)) = sa1;
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(B) + lds_off_bf16(st_row, st_k8)) = sb0;
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(B) + lds_off_bf16(st_row + 64, st_k8)) = sb1;
     __syncthreads();
 
     // ---- issue next tile's loads (consumed at the next ds_write) ----
-    if (step + 1 < nsteps) load_step(step + 1, na, nb);
+    if (step + 1 < nsteps) RTHD_LOAD_STEP(step + 1, na0, na1, nb0, nb1);
 
     // ---- fragments + MFMA ----
     const int arow_base = wr * 64 + (lane & 15);
@@ -230,12 +249,12 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
 
     __syncthreads();
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      sa[h] = na[h];
-      sb[h] = nb[h];
-    }
+    sa0 = na0;
+    sa1 = na1;
+    sb0 = nb0;
+    sb1 = nb1;
   }
+#undef RTHD_LOAD_STEP
 
   // ---- epilogue ----
   const int col0 = nblk * 128 + wc * 64 + (lane & 15);

@@ -167,24 +167,31 @@ void stem_wgrad_kernel(const T* __restrict__ x, const T* __restrict__ dy,
     }
   }
 
-  // reduce the 4 streams through LDS, then atomics into dw
-  __shared__ float sh[SW_STREAMS][64][KS * STEM_CIN];
+  // reduce the 4 streams sequentially through a SMALL LDS tile (a
+  // [4][64][21] buffer was 86 KB -> 1 block/CU), then atomics into dw
+  __shared__ float sh[64][KS * STEM_CIN];
+  for (int sstep = 0; sstep < SW_STREAMS; ++sstep) {
+    if (stream == sstep) {
 #pragma unroll
-  for (int tx = 0; tx < KS; ++tx)
+      for (int tx = 0; tx < KS; ++tx)
 #pragma unroll
-    for (int ci = 0; ci < STEM_CIN; ++ci)
-      sh[stream][co][tx * STEM_CIN + ci] = acc[tx][ci];
-  __syncthreads();
+        for (int ci = 0; ci < STEM_CIN; ++ci) {
+          const int e = tx * STEM_CIN + ci;
+          if (sstep == 0)
+            sh[co][e] = acc[tx][ci];
+          else
+            sh[co][e] += acc[tx][ci];
+        }
+    }
+    __syncthreads();
+  }
   if (stream == 0) {
 #pragma unroll
     for (int tx = 0; tx < KS; ++tx)
 #pragma unroll
-      for (int ci = 0; ci < STEM_CIN; ++ci) {
-        const int e = tx * STEM_CIN + ci;
-        const float v = sh[0][co][e] + sh[1][co][e] + sh[2][co][e] +
-                        sh[3][co][e];
-        atomicAdd(&dw[((co * STEM_CIN + ci) * KS + ty) * KS + tx], v);
-      }
+      for (int ci = 0; ci < STEM_CIN; ++ci)
+        atomicAdd(&dw[((co * STEM_CIN + ci) * KS + ty) * KS + tx],
+                  sh[co][tx * STEM_CIN + ci]);
   }
 }
 
@@ -198,7 +205,7 @@ torch::Tensor stem_wgrad(torch::Tensor x, torch::Tensor dy, int64_t stride,
   const int M = B * Ho * Wo;
   auto dw = torch::zeros({STEM_COUT, STEM_CIN, 7, 7},
                          xc.options().dtype(at::kFloat));
-  int chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 4096, 1), 256);
+  int chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 16384, 1), 64);
   int chunk_len = (int)cdiv(M, chunks);
   chunks = (int)cdiv(M, chunk_len);
   dim3 grid(7, chunks);

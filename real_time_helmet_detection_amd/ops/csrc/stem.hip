@@ -205,7 +205,7 @@ torch::Tensor stem_wgrad(torch::Tensor x, torch::Tensor dy, int64_t stride,
   const int M = B * Ho * Wo;
   auto dw = torch::zeros({STEM_COUT, STEM_CIN, 7, 7},
                          xc.options().dtype(at::kFloat));
-  int chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 16384, 1), 64);
+  int chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 4096, 1), 128);
   int chunk_len = (int)cdiv(M, chunks);
   chunks = (int)cdiv(M, chunk_len);
   dim3 grid(7, chunks);

@@ -109,7 +109,7 @@ struct ConvGeo {
 };
 
 // act codes from common.h; epilogue: y = act(acc*scale[c] + shift[c] (+skip))
-template <bool HAS_SKIP>
+template <bool HAS_SKIP, bool ALIGNED>
 __global__ __launch_bounds__(256)
 void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ wpk,
@@ -170,33 +170,41 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
     {                                                                        \
       const int iy_ = ay[0] * g.stride + dy_;                                \
       const int ix_ = ax[0] * g.stride + dx_;                                \
-      if (am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
-          c0_ < g.Cin) {                                                     \
+      const bool val_ = am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 &&  \
+                        ix_ < g.W && c0_ < g.Cin;                            \
+      if (ALIGNED) {                                                         \
+        /* unconditional clamped load + select-zero: a branch around the  */ \
+        /* load costs a vmcnt(0) drain per element (guide trap (c))       */ \
+        const int64_t o_ = val_                                              \
+            ? (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin + c0_       \
+            : 0;                                                             \
+        v0_ = *reinterpret_cast<const uint4*>(x + o_);                       \
+        if (!val_) v0_ = uint4{0, 0, 0, 0};                                  \
+      } else if (val_) {                                                     \
         const bf16* src_ =                                                   \
             x + (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
-        if (c0_ + 8 <= g.Cin) {                                              \
-          v0_ = *reinterpret_cast<const uint4*>(src_);                       \
-        } else {                                                             \
-          bf16 tmp_[8] = {};                                                 \
-          for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];      \
-          v0_ = *reinterpret_cast<const uint4*>(tmp_);                       \
-        }                                                                    \
+        bf16 tmp_[8] = {};                                                   \
+        for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];        \
+        v0_ = *reinterpret_cast<const uint4*>(tmp_);                         \
       }                                                                      \
     }                                                                        \
     {                                                                        \
       const int iy_ = ay[1] * g.stride + dy_;                                \
       const int ix_ = ax[1] * g.stride + dx_;                                \
-      if (am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
-          c0_ < g.Cin) {                                                     \
+      const bool val_ = am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 &&  \
+                        ix_ < g.W && c0_ < g.Cin;                            \
+      if (ALIGNED) {                                                         \
+        const int64_t o_ = val_                                              \
+            ? (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin + c0_       \
+            : 0;                                                             \
+        v1_ = *reinterpret_cast<const uint4*>(x + o_);                       \
+        if (!val_) v1_ = uint4{0, 0, 0, 0};                                  \
+      } else if (val_) {                                                     \
         const bf16* src_ =                                                   \
             x + (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
-        if (c0_ + 8 <= g.Cin) {                                              \
-          v1_ = *reinterpret_cast<const uint4*>(src_);                       \
-        } else {                                                             \
-          bf16 tmp_[8] = {};                                                 \
-          for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];      \
-          v1_ = *reinterpret_cast<const uint4*>(tmp_);                       \
-        }                                                                    \
+        bf16 tmp_[8] = {};                                                   \
+        for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];        \
+        v1_ = *reinterpret_cast<const uint4*>(tmp_);                         \
       }                                                                      \
     }                                                                        \
     (A0) = v0_;                                                              \
@@ -464,12 +472,21 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
     auto* py = reinterpret_cast<bf16*>(y.data_ptr());
     const bf16* ps =
         has_skip ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
-    if (has_skip)
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true>), grid,
+    const bool aligned = g.Cin % 8 == 0;
+    if (has_skip && aligned)
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true, true>), grid,
+          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
+          sh.data_ptr<float>(), ps, py, g, (int)act);
+    else if (has_skip)
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true, false>), grid,
+          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
+          sh.data_ptr<float>(), ps, py, g, (int)act);
+    else if (aligned)
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false, true>), grid,
           dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
           sh.data_ptr<float>(), ps, py, g, (int)act);
     else
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false>), grid,
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false, false>), grid,
           dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
           sh.data_ptr<float>(), ps, py, g, (int)act);
   } else {

@@ -47,13 +47,18 @@ void stem_fwd_kernel(const T* __restrict__ x,
 #pragma unroll 1
     for (int ty = 0; ty < KS; ++ty) {
       const int iy = oy * stride + ty - pad;
-      if (iy < 0 || iy >= H) continue;
+      const bool row_ok = iy >= 0 && iy < H;
 #pragma unroll 1
       for (int tx = 0; tx < KS; ++tx) {
         const int ix = ox * stride + tx - pad;
-        if (ix < 0 || ix >= W) continue;
-        const T* px = x + (((int64_t)b * H + iy) * W + ix) * STEM_CIN;
-        const float i0 = ldf(&px[0]), i1 = ldf(&px[1]), i2 = ldf(&px[2]);
+        // clamped unconditional loads + select-zero: a branch around the
+        // loads makes hipcc drain vmcnt(0) per tap (guide §5 trap (c))
+        const bool val = row_ok && ix >= 0 && ix < W;
+        const int64_t off = val
+            ? (((int64_t)b * H + iy) * W + ix) * STEM_CIN : 0;
+        const T* px = x + off;
+        float i0 = ldf(&px[0]), i1 = ldf(&px[1]), i2 = ldf(&px[2]);
+        if (!val) { i0 = 0.f; i1 = 0.f; i2 = 0.f; }
         const float* wt = wl + (ty * KS + tx) * STEM_CIN * STEM_COUT;
 #pragma unroll
         for (int co = 0; co < STEM_COUT; ++co) {
@@ -144,20 +149,24 @@ void stem_wgrad_kernel(const T* __restrict__ x, const T* __restrict__ dy,
 
   for (; m < px1; m += SW_STREAMS) {
     const int iy = oy * stride + ty - pad;
-    if (iy >= 0 && iy < H) {
-      const float dyv = ldf(&dy[(int64_t)m * STEM_COUT + co]);
-      if (dyv != 0.f) {
-        const T* xrow = x + (((int64_t)b * H + iy) * W) * STEM_CIN;
+    const bool row_ok = iy >= 0 && iy < H;
+    const float dyv = ldf(&dy[(int64_t)m * STEM_COUT + co]);
+    const T* xrow =
+        x + (((int64_t)b * H + (row_ok ? iy : 0)) * W) * STEM_CIN;
 #pragma unroll
-        for (int tx = 0; tx < KS; ++tx) {
-          const int ix = ox * stride + tx - pad;
-          if (ix < 0 || ix >= W) continue;
+    for (int tx = 0; tx < KS; ++tx) {
+      const int ix = ox * stride + tx - pad;
+      // clamped unconditional loads + select (guide §5 trap (c))
+      const bool val = row_ok && ix >= 0 && ix < W;
+      const int ixs = val ? ix : 0;
+      float xv[STEM_CIN];
 #pragma unroll
-          for (int ci = 0; ci < STEM_CIN; ++ci)
-            acc[tx][ci] = fmaf(ldf(&xrow[ix * STEM_CIN + ci]), dyv,
-                               acc[tx][ci]);
-        }
-      }
+      for (int ci = 0; ci < STEM_CIN; ++ci)
+        xv[ci] = ldf(&xrow[ixs * STEM_CIN + ci]);
+      const float d = val ? dyv : 0.f;
+#pragma unroll
+      for (int ci = 0; ci < STEM_CIN; ++ci)
+        acc[tx][ci] = fmaf(xv[ci], d, acc[tx][ci]);
     }
     // advance decomposition by SW_STREAMS pixels
     ox += SW_STREAMS;

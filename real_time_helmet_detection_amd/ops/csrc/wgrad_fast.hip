@@ -35,6 +35,7 @@ struct WgradGeo2 {
   int chunk_len;
 };
 
+template <bool ALIGNED>  // Cin,Cout % 8 == 0: branch-free staging loads
 __global__ __launch_bounds__(512)
 void wgrad_bf16_kernel(const bf16* __restrict__ x,
                        const bf16* __restrict__ dy,
@@ -81,25 +82,32 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
     ushort r[8][8];
     if (stage_x) {
       int bj = bb, oyj = oy, oxj = ox;
+      const int cbase = ci0 + s_ci;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int m = p0 + s_px + j;
-        uint4 v = {0, 0, 0, 0};
-        if (m < px_end) {
-          const int iy = oyj * g.stride + dyt;
-          const int ix = oxj * g.stride + dxt;
-          if (iy >= 0 && iy < g.H && ix >= 0 && ix < g.W) {
-            const int cbase = ci0 + s_ci;
-            if (cbase + 8 <= g.Cin) {
-              v = *reinterpret_cast<const uint4*>(
-                  x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
-            } else if (cbase < g.Cin) {
-              const ushort* src = reinterpret_cast<const ushort*>(
-                  x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
-              ushort tmp[8] = {};
-              for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
-              v = *reinterpret_cast<const uint4*>(tmp);
-            }
+        uint4 v;
+        const int iy = oyj * g.stride + dyt;
+        const int ix = oxj * g.stride + dxt;
+        const bool val = m < px_end && iy >= 0 && iy < g.H && ix >= 0 &&
+                         ix < g.W && cbase < g.Cin;
+        if (ALIGNED) {
+          // UNCONDITIONAL load from a clamped address + select-zero: a
+          // branch around the load makes hipcc drain vmcnt(0) per element
+          // (guide §5 trap (c) — measured 2-6x on this kernel)
+          const int64_t off = val
+              ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase
+              : 0;
+          v = *reinterpret_cast<const uint4*>(x + off);
+          if (!val) v = uint4{0, 0, 0, 0};
+        } else {
+          v = uint4{0, 0, 0, 0};
+          if (val) {
+            const ushort* src = reinterpret_cast<const ushort*>(
+                x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
+            ushort tmp[8] = {};
+            for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
+            v = *reinterpret_cast<const uint4*>(tmp);
           }
         }
         *reinterpret_cast<uint4*>(r[j]) = v;
@@ -118,16 +126,19 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
             wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
       }
     } else {
+      const int cbase = co0 + s_ci;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int m = p0 + s_px + j;
-        uint4 v = {0, 0, 0, 0};
-        if (m < px_end) {
-          const int cbase = co0 + s_ci;
-          if (cbase + 8 <= g.Cout) {
-            v = *reinterpret_cast<const uint4*>(
-                dy + (int64_t)m * g.Cout + cbase);
-          } else if (cbase < g.Cout) {
+        uint4 v;
+        const bool val = m < px_end && cbase < g.Cout;
+        if (ALIGNED) {
+          const int64_t off = val ? (int64_t)m * g.Cout + cbase : 0;
+          v = *reinterpret_cast<const uint4*>(dy + off);
+          if (!val) v = uint4{0, 0, 0, 0};
+        } else {
+          v = uint4{0, 0, 0, 0};
+          if (val) {
             const ushort* src =
                 reinterpret_cast<const ushort*>(dy + (int64_t)m * g.Cout +
                                                 cbase);
@@ -229,10 +240,17 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
 
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(wgrad_bf16_kernel, grid, dim3(512), 0, s,
-      reinterpret_cast<const bf16*>(xc.data_ptr()),
-      reinterpret_cast<const bf16*>(dyc.data_ptr()),
-      dw.data_ptr<float>(), g);
+  const bool aligned = (g.Cin % 8 == 0) && (g.Cout % 8 == 0);
+  if (aligned)
+    hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(512), 0, s,
+        reinterpret_cast<const bf16*>(xc.data_ptr()),
+        reinterpret_cast<const bf16*>(dyc.data_ptr()),
+        dw.data_ptr<float>(), g);
+  else
+    hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(512), 0, s,
+        reinterpret_cast<const bf16*>(xc.data_ptr()),
+        reinterpret_cast<const bf16*>(dyc.data_ptr()),
+        dw.data_ptr<float>(), g);
   HIP_CHECK_LAST();
   return dw;
 }

@@ -36,38 +36,39 @@ struct WgradGeo2 {
 };
 
 template <bool ALIGNED>  // Cin,Cout % 8 == 0: branch-free staging loads
-__global__ __launch_bounds__(512)
+__global__ __launch_bounds__(256)
 void wgrad_bf16_kernel(const bf16* __restrict__ x,
                        const bf16* __restrict__ dy,
                        float* __restrict__ dw, WgradGeo2 g) {
-  // 8 waves; block tile [128 ci][128 co] (wave = 32 ci x 64 co) — twice the
-  // MFMA work per staged LDS byte of the 4-wave 64x64 version.
+  // 4 waves; block tile [64 ci][64 co] (wave = 32 x 32). A 128x128 8-wave
+  // variant measured 2.2x SLOWER per unit work (64 KB LDS -> 2 blocks/CU)
+  // — this geometry keeps 5 blocks/CU resident.
   const int t = blockIdx.z % (g.KH * g.KW);
   const int chunk = blockIdx.z / (g.KH * g.KW);
-  const int ci0 = blockIdx.x * 128;
-  const int co0 = blockIdx.y * 128;
+  const int ci0 = blockIdx.x * 64;
+  const int co0 = blockIdx.y * 64;
   const int dyt = t / g.KW - g.pad;
   const int dxt = t % g.KW - g.pad;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;             // 8 waves: 4 (ci) x 2 (co)
+  const int wid = tid >> 6;             // 4 waves: 2 (ci) x 2 (co)
   const int wr = wid >> 1, wc = wid & 1;
 
-  __shared__ __attribute__((aligned(16))) char smem[2 * 128 * 256];
-  bf16* Xl = reinterpret_cast<bf16*>(smem);           // 32 KB
-  bf16* Yl = reinterpret_cast<bf16*>(smem + 32768);   // 32 KB
+  __shared__ __attribute__((aligned(16))) char smem[2 * 64 * 256];
+  bf16* Xl = reinterpret_cast<bf16*>(smem);           // 16 KB
+  bf16* Yl = reinterpret_cast<bf16*>(smem + 16384);   // 16 KB
 
-  f32x4 acc[2][4] = {};
+  f32x4 acc[2][2] = {};
 
   const int px_start = chunk * g.chunk_len;
   const int px_end = min(px_start + g.chunk_len, g.M);
 
-  // staging role: half 0 -> X, half 1 -> dY; within a half (256 threads):
-  // ch octet = tid & 15 (16 octs = 128 ch), px block = (tid >> 4) & 15
-  const bool stage_x = tid < 256;
-  const int s_ci = (tid & 15) * 8;
-  const int s_px = ((tid >> 4) & 15) * 8;
+  // staging role: half 0 -> X, half 1 -> dY; within a half (128 threads):
+  // ch octet = tid & 7 (8 octs = 64 ch), px block = (tid >> 3) & 15
+  const bool stage_x = tid < 128;
+  const int s_ci = (tid & 7) * 8;
+  const int s_px = ((tid >> 3) & 15) * 8;
 
   // incremental (b, oy, ox) for this thread's px base (X half only)
   const int HoWo = g.Ho * g.Wo;
@@ -172,23 +173,20 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       const int k8 = (lane >> 4) + ks * 4;
-      bf16x8 xa[2], yb[4];
+      bf16x8 xa[2], yb[2];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
         const int arow = wr * 32 + i * 16 + (lane & 15);
         xa[i] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<char*>(Xl) + wg_off(arow, k8));
-      }
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int brow = wc * 64 + i * 16 + (lane & 15);
+        const int brow = wc * 32 + i * 16 + (lane & 15);
         yb[i] = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<char*>(Yl) + wg_off(brow, k8));
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < 2; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               xa[mi], yb[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -202,8 +200,8 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
       const int ci = ci0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r2;
       if (ci >= g.Cin) continue;
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const int co = co0 + wc * 64 + ni * 16 + (lane & 15);
+      for (int ni = 0; ni < 2; ++ni) {
+        const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
         atomicAdd(&dw[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx],
                   acc[mi][ni][r2]);
@@ -229,8 +227,8 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
 
   auto dw = torch::zeros({g.Cout, g.Cin, KH, KW},
                          xc.options().dtype(at::kFloat));
-  const int ci_tiles = (int)cdiv(g.Cin, 128);
-  const int co_tiles = (int)cdiv(g.Cout, 128);
+  const int ci_tiles = (int)cdiv(g.Cin, 64);
+  const int co_tiles = (int)cdiv(g.Cout, 64);
   const int taps = (int)(KH * KW);
   int nchunks = std::max(1, 1024 / (ci_tiles * co_tiles * taps));
   int chunk_len = (int)cdiv(g.M, nchunks);
@@ -242,12 +240,12 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   auto s = at::cuda::getCurrentCUDAStream();
   const bool aligned = (g.Cin % 8 == 0) && (g.Cout % 8 == 0);
   if (aligned)
-    hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(512), 0, s,
+    hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(256), 0, s,
         reinterpret_cast<const bf16*>(xc.data_ptr()),
         reinterpret_cast<const bf16*>(dyc.data_ptr()),
         dw.data_ptr<float>(), g);
   else
-    hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(512), 0, s,
+    hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(256), 0, s,
         reinterpret_cast<const bf16*>(xc.data_ptr()),
         reinterpret_cast<const bf16*>(dyc.data_ptr()),
         dw.data_ptr<float>(), g);

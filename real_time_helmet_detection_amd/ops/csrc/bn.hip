@@ -36,8 +36,9 @@ static bool fast8_ok(const torch::Tensor& t, int C) {
 }
 
 static int pick_chunks(int64_t M, int C) {
+  // enough blocks to fill 256 CUs (a 256 cap left 1 block/CU -> 1.5 TB/s)
   int64_t chunks = cdiv((int64_t)M * C, (int64_t)16384);
-  if (chunks > 256) chunks = 256;
+  if (chunks > 1024) chunks = 1024;
   if (chunks < 1) chunks = 1;
   return (int)chunks;
 }
@@ -136,6 +137,8 @@ __global__ void colsum8_part_kernel(const bf16* __restrict__ x,
 // out[c] = sum_k p[k][c]  (and out2/p2 when given).
 // block = 64 channels x 4 k-substreams (a K=256 serial loop in one tiny
 // block was 44 us — latency-bound).
+// grid (C/64, KSPLIT): each block sums its K-slice, atomically adds into
+// the zero-initialized outputs (KSPLIT same-address atomics: negligible)
 __global__ void reduce_partials_kernel(const float* __restrict__ p1,
                                        const float* __restrict__ p2,
                                        float* __restrict__ o1,
@@ -143,9 +146,11 @@ __global__ void reduce_partials_kernel(const float* __restrict__ p1,
                                        int K, int C) {
   const int c = blockIdx.x * 64 + (threadIdx.x & 63);
   const int sub = threadIdx.x >> 6;
+  const int k0 = (K * blockIdx.y) / gridDim.y;
+  const int k1 = (K * (blockIdx.y + 1)) / gridDim.y;
   float a = 0.f, b = 0.f;
   if (c < C) {
-    for (int k = sub; k < K; k += 4) {
+    for (int k = k0 + sub; k < k1; k += 4) {
       a += p1[(int64_t)k * C + c];
       if (p2) b += p2[(int64_t)k * C + c];
     }
@@ -157,11 +162,11 @@ __global__ void reduce_partials_kernel(const float* __restrict__ p1,
   if (sub == 0 && c < C) {
     a = sha[threadIdx.x] + sha[threadIdx.x + 64] + sha[threadIdx.x + 128] +
         sha[threadIdx.x + 192];
-    o1[c] = a;
+    atomicAdd(&o1[c], a);
     if (p2) {
       b = shb[threadIdx.x] + shb[threadIdx.x + 64] + shb[threadIdx.x + 128] +
           shb[threadIdx.x + 192];
-      o2[c] = b;
+      atomicAdd(&o2[c], b);
     }
   }
 }
@@ -449,7 +454,8 @@ static void run_colsum(const torch::Tensor& xc, torch::Tensor& sum,
             p1.data_ptr<float>(), p2p, M, C);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64)), dim3(256),
+  hipLaunchKernelGGL(reduce_partials_kernel,
+      dim3(cdiv(C, 64), std::min(8, chunks)), dim3(256),
       0, s, p1.data_ptr<float>(), p2p, sum.data_ptr<float>(),
       sumsq ? sumsq->data_ptr<float>() : nullptr, chunks, C);
 }
@@ -461,8 +467,8 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
   const int64_t M = xc.numel() / C;
-  auto sum = torch::empty({C}, xc.options().dtype(at::kFloat));
-  auto sumsq = torch::empty({C}, xc.options().dtype(at::kFloat));
+  auto sum = torch::zeros({C}, xc.options().dtype(at::kFloat));
+  auto sumsq = torch::zeros({C}, xc.options().dtype(at::kFloat));
   auto mean = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto rstd = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
@@ -524,8 +530,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   const int C = xc.size(1);
   const int64_t n = xc.numel();
   const int64_t M = n / C;
-  auto s1 = torch::empty({C}, xc.options().dtype(at::kFloat));
-  auto s2 = torch::empty({C}, xc.options().dtype(at::kFloat));
+  auto s1 = torch::zeros({C}, xc.options().dtype(at::kFloat));
+  auto s2 = torch::zeros({C}, xc.options().dtype(at::kFloat));
   auto dx = torch::empty_like(xc);
   auto gm = gamma.to(at::kFloat).contiguous();
   auto bt = beta.to(at::kFloat).contiguous();
@@ -553,7 +559,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
           p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64)), dim3(256),
+  hipLaunchKernelGGL(reduce_partials_kernel,
+      dim3(cdiv(C, 64), std::min(8, chunks)), dim3(256),
       0, s, p1.data_ptr<float>(), p2.data_ptr<float>(),
       s1.data_ptr<float>(), s2.data_ptr<float>(), chunks, C);
 
@@ -590,7 +597,7 @@ torch::Tensor col_sum(torch::Tensor x) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
   const int64_t M = xc.numel() / C;
-  auto sum = torch::empty({C}, xc.options().dtype(at::kFloat));
+  auto sum = torch::zeros({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
   run_colsum(xc, sum, nullptr, M, C, s);
   HIP_CHECK_LAST();

@@ -22,10 +22,14 @@ namespace rthd {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-// [64 rows][128 px] bf16: row stride 256 B = 16 slots of 16 B;
-// slot' = slot ^ (row & 15) -> conflict-free 16-lane fragment reads.
+// [64 rows][128 px] bf16 with ONE SLOT of row padding (272 B rows).
+// bank(addr) = (17*row + 4*slot) % 64: distinct for the write groups
+// (rows e+8k, fixed slot: 8k mod 64 distinct) AND the fragment-read groups
+// (rows 0..15, fixed slot: 17*row mod 64 distinct). The earlier XOR slot
+// swizzle on 256-B rows left 4-way WRITE conflicts (row*64 = 0 mod 64
+// erases the row term) — measured 4.4e9 conflict cycles per kbench run.
 DEV_INLINE int wg_off(int row, int px8) {
-  return row * 256 + ((px8 ^ (row & 15)) << 4);
+  return row * 272 + (px8 << 4);
 }
 
 struct WgradGeo2 {
@@ -55,9 +59,9 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   const int wid = tid >> 6;             // 4 waves: 2 (ci) x 2 (co)
   const int wr = wid >> 1, wc = wid & 1;
 
-  __shared__ __attribute__((aligned(16))) char smem[2 * 64 * 256];
-  bf16* Xl = reinterpret_cast<bf16*>(smem);           // 16 KB
-  bf16* Yl = reinterpret_cast<bf16*>(smem + 16384);   // 16 KB
+  __shared__ __attribute__((aligned(16))) char smem[2 * 64 * 272];
+  bf16* Xl = reinterpret_cast<bf16*>(smem);           // 17 KB (padded rows)
+  bf16* Yl = reinterpret_cast<bf16*>(smem + 64 * 272);
 
   f32x4 acc[2][2] = {};
 

@@ -28,8 +28,11 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // (rows 0..15, fixed slot: 17*row mod 64 distinct). The earlier XOR slot
 // swizzle on 256-B rows left 4-way WRITE conflicts (row*64 = 0 mod 64
 // erases the row term) — measured 4.4e9 conflict cycles per kbench run.
+// slot' = px8 ^ (row>>3): write groups (8 lanes, rows e+8L, fixed px) get
+// distinct banks (32L parity + 4*(px^L) spread); read groups (rows 16a..+15,
+// fixed k8) stay ~conflict-free on the padded 272-B rows.
 DEV_INLINE int wg_off(int row, int px8) {
-  return row * 272 + (px8 << 4);
+  return row * 272 + ((px8 ^ ((row >> 3) & 7)) << 4);
 }
 
 struct WgradGeo2 {

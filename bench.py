@@ -121,7 +121,10 @@ def main():
     is_bucketed = world > 1
 
     if args.mode == 'train':
-        opt = torch.optim.Adam(net.parameters(), lr=5e-4)
+        try:
+            opt = torch.optim.Adam(net.parameters(), lr=5e-4, fused=True)
+        except (RuntimeError, TypeError, ValueError):
+            opt = torch.optim.Adam(net.parameters(), lr=5e-4, foreach=True)
 
         def step(i):
             img, hm, off, wh, mask = batches[i % len(batches)]

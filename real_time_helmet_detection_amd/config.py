@@ -212,6 +212,10 @@ def get_arguments(argv=None):
         os.environ.setdefault('CUDA_DEVICE_ORDER', 'PCI_BUS_ID')
         os.environ['CUDA_VISIBLE_DEVICES'] = ','.join(map(str, args.gpu_no))
 
+    if getattr(args, 'engine', 'auto') == 'torch':
+        # explicit eager-torch engine on GPU (A/B measurement only)
+        os.environ['RTHD_EAGER_GPU'] = '1'
+
     with open(os.path.join(args.save_path, 'argument.txt'), 'w') as f:
         for key, value in sorted(vars(args).items()):
             f.write('%s: %s' % (key, value) + '\n')

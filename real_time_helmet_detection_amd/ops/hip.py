@@ -124,16 +124,31 @@ class _ConvBNActFn(torch.autograd.Function):
                 raise NotImplementedError(
                     'stem dgrad (3-channel input) is not needed: the image '
                     'is a leaf tensor')
-            if stride != 1:
-                raise NotImplementedError(
-                    'dgrad for stride>1 convs (pool="Conv") is not '
-                    'implemented in the HIP engine yet')
             cin = xc.shape[1]
-            wpk_t = C.pack_weights(weight, True, bf16)
             ones = torch.ones(cin, device=xc.device, dtype=torch.float32)
             zeros = torch.zeros(cin, device=xc.device, dtype=torch.float32)
-            dx = C.conv_fwd(dpre, wpk_t, ones, zeros, None, kh, kw, 1, pad,
-                            cin, ACT_CODE['Linear'])
+            if stride == 1:
+                wpk_t = C.pack_weights(weight, True, bf16)
+                dx = C.conv_fwd(dpre, wpk_t, ones, zeros, None, kh, kw, 1,
+                                pad, cin, ACT_CODE['Linear'])
+            elif stride == 2 and kh == 2 and kw == 2 and pad == 0:
+                # k2/s2 windows don't overlap: each input pixel receives
+                # from exactly one output pixel, so dgrad decomposes into
+                # FOUR 1x1 convs (one per tap parity), scattered back into
+                # the interleaved input grid (the pool='Conv' downsampler).
+                dx = torch.empty_like(xc).contiguous(
+                    memory_format=torch.channels_last)
+                for r in range(2):
+                    for c in range(2):
+                        wt = weight[:, :, r:r + 1, c:c + 1]
+                        wpk_t = C.pack_weights(wt, True, bf16)
+                        part = C.conv_fwd(dpre, wpk_t, ones, zeros, None,
+                                          1, 1, 1, 0, cin,
+                                          ACT_CODE['Linear'])
+                        dx[:, :, r::2, c::2] = part
+            else:
+                raise NotImplementedError(
+                    f'HIP dgrad for k={kh} stride={stride} not implemented')
 
         # wgrad
         dw = None

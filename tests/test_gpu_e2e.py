@@ -161,3 +161,23 @@ def test_sub_division_accumulation_gpu():
         # BN batch stats differ between full batch and micro-batches, so
         # tolerances are loose; this checks accumulation plumbing, not BN.
         assert torch.isfinite(pb.grad).all()
+
+
+def test_conv_pool_backward_gpu():
+    """pool='Conv' (k2 s2) backward: stride-2 dgrad via tap decomposition."""
+    from real_time_helmet_detection_amd.models import Pool
+    import torch.nn.functional as F
+    torch.manual_seed(6)
+    p_cpu = Pool(32, 'Conv')
+    p_gpu = copy.deepcopy(p_cpu).cuda().to(memory_format=CL)
+    x = torch.randn(2, 32, 16, 16)
+    xg = x.cuda().contiguous(memory_format=CL).requires_grad_(True)
+    xc = x.clone().requires_grad_(True)
+    y_cpu = p_cpu(xc)
+    y_gpu = p_gpu(xg)
+    dy = torch.randn_like(y_cpu)
+    y_cpu.backward(dy)
+    y_gpu.backward(dy.cuda().contiguous(memory_format=CL))
+    assert rel_err(y_gpu, y_cpu) < 1e-4
+    assert rel_err(xg.grad, xc.grad) < 1e-4
+    assert rel_err(p_gpu.pool.weight.grad, p_cpu.pool.weight.grad) < 1e-3

@@ -71,3 +71,28 @@ class GradScaler:
 
     def is_enabled(self):
         return self._enabled
+
+
+# ---------------------------- fp8 inference mode ----------------------------
+# BASELINE config 5: fp8 e4m3 MFMA inference. Weights are per-channel scaled
+# into e4m3 range (scale folded into the conv epilogue); activations are
+# converted bf16 -> fp8 inside the conv staging (saturating). Inference only.
+
+_fp8_depth = 0
+
+
+def fp8_enabled():
+    return _fp8_depth > 0
+
+
+@contextlib.contextmanager
+def fp8_autocast(enabled=True):
+    global _fp8_depth
+    if not enabled:
+        yield
+        return
+    _fp8_depth += 1
+    try:
+        yield
+    finally:
+        _fp8_depth -= 1

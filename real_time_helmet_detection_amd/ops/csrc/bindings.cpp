@@ -36,6 +36,12 @@ torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
                       double iou_threshold);
 
 torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16);
+torch::Tensor pack_weights_fp8(torch::Tensor w);
+torch::Tensor conv_fwd_fp8(torch::Tensor x, torch::Tensor wpk,
+                           torch::Tensor scale, torch::Tensor shift,
+                           c10::optional<torch::Tensor> skip,
+                           int64_t KH, int64_t KW, int64_t stride,
+                           int64_t pad, int64_t Cout, int64_t act);
 torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                        torch::Tensor scale, torch::Tensor shift,
                        c10::optional<torch::Tensor> skip,
@@ -79,6 +85,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_fwd", &rthd::decode_fwd);
   m.def("nms_fwd", &rthd::nms_fwd);
   m.def("pack_weights", &rthd::pack_weights);
+  m.def("pack_weights_fp8", &rthd::pack_weights_fp8);
+  m.def("conv_fwd_fp8", &rthd::conv_fwd_fp8);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("stem_fwd", &rthd::stem_fwd);
   m.def("stem_wgrad", &rthd::stem_wgrad);

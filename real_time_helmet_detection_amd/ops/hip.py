@@ -164,6 +164,37 @@ class _ConvBNActFn(torch.autograd.Function):
                 None, None, None, None, None, None, None, None, None, None)
 
 
+def _conv_infer_fp8(x, conv, bn, act_code):
+    """Inference-only fp8 path: per-cout-scaled e4m3 weights (cached on the
+    module keyed by weight version), activations converted in staging."""
+    C = _C()
+    kh, kw = conv.kernel_size
+    stride, pad = conv.stride[0], conv.padding[0]
+    cout = conv.weight.shape[0]
+    cache = getattr(conv, '_rthd_fp8_cache', None)
+    ver = conv.weight._version
+    if cache is None or cache[0] != ver:
+        w = conv.weight.detach().float()
+        sw = w.abs().amax(dim=(1, 2, 3)).clamp(min=1e-8) / 240.0
+        wpk = C.pack_weights_fp8(w / sw.view(-1, 1, 1, 1))
+        conv._rthd_fp8_cache = (ver, wpk, sw)
+        cache = conv._rthd_fp8_cache
+    _, wpk, sw = cache
+
+    if bn is not None:
+        rstd_run = torch.rsqrt(bn.running_var.float() + bn.eps)
+        scale = (bn.weight.float() * rstd_run) * sw
+        bias_f = (conv.bias.float() if conv.bias is not None
+                  else torch.zeros_like(sw))
+        shift = bn.bias.float() + (bias_f - bn.running_mean.float())             * bn.weight.float() * rstd_run
+    else:
+        scale = sw
+        shift = (conv.bias.float() if conv.bias is not None
+                 else torch.zeros(cout, device=x.device))
+    return C.conv_fwd_fp8(x, wpk, scale.contiguous(), shift.contiguous(),
+                          None, kh, kw, stride, pad, cout, act_code)
+
+
 def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
     """GPU twin of functional.conv_bn_act (act_module path stays eager)."""
     kh, kw = conv.kernel_size
@@ -178,6 +209,12 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
     act_code = ACT_CODE.get(act)
     if act_code is None:
         raise NotImplementedError(f'HIP conv epilogue: activation {act!r}')
+    if (_amp.fp8_enabled() and not training and not is_stem
+            and not torch.is_grad_enabled() and cin % 16 == 0):
+        y = _conv_infer_fp8(x, conv, bn, act_code)
+        if act_module is not None:
+            y = act_module(y)
+        return y
     y = _ConvBNActFn.apply(
         x, conv.weight,
         conv.bias,

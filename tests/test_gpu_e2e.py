@@ -181,3 +181,38 @@ def test_conv_pool_backward_gpu():
     assert rel_err(y_gpu, y_cpu) < 1e-4
     assert rel_err(xg.grad, xc.grad) < 1e-4
     assert rel_err(p_gpu.pool.weight.grad, p_cpu.pool.weight.grad) < 1e-3
+
+
+def test_fp8_inference_close_to_bf16():
+    from real_time_helmet_detection_amd import amp
+    torch.manual_seed(7)
+    _, gpu = _models(in_ch=64, seed=7)
+    gpu.eval()
+    x = torch.randn(2, 3, 64, 64).cuda().contiguous(memory_format=CL)
+    with torch.no_grad():
+        with amp.autocast(True):
+            ref = gpu(x)
+        with amp.autocast(True), amp.fp8_autocast(True):
+            got = gpu(x)
+    # e4m3 weights+activations through ~25 layers: loose tolerance
+    assert rel_err(got, ref) < 0.25
+    assert torch.isfinite(got).all()
+
+
+def test_graphed_predictor_matches_eager():
+    from real_time_helmet_detection_amd.engine.evaluator import (
+        Prediction, GraphedPredictor)
+    from real_time_helmet_detection_amd.models import StackedHourglass
+    torch.manual_seed(8)
+    net = StackedHourglass(1, 32, 6).cuda().to(memory_format=CL).eval()
+    pred = Prediction(net, topk=20, scale_factor=4, conf_th=0.1, nms='nms',
+                      nms_th=0.5).cuda()
+    x = torch.randn(2, 3, 128, 128).cuda().contiguous(memory_format=CL)
+    g = GraphedPredictor(pred, x.clone())
+    from real_time_helmet_detection_amd import amp
+    with torch.no_grad(), amp.autocast(True):
+        want = pred(x)
+    got = g(x)
+    for wb, gb in zip(want[0], got[0]):
+        assert wb.shape == gb.shape
+        torch.testing.assert_close(gb, wb, rtol=0.05, atol=0.5)

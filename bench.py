@@ -46,6 +46,8 @@ def build_args():
                         'torch-ROCm ops (A/B baseline only)')
     p.add_argument('--fp8', action='store_true', default=False,
                    help='fp8 e4m3 MFMA inference (infer mode only)')
+    p.add_argument('--no-train-graph', action='store_true', default=False,
+                   help='disable hipGraph capture of the training step')
     p.add_argument('--graph', action='store_true', default=False,
                    help='hipGraph-captured inference (infer mode only)')
     p.add_argument('--channels-last', dest='channels_last',
@@ -125,12 +127,15 @@ def main():
     is_bucketed = world > 1
 
     if args.mode == 'train':
+        want_graph = use_cuda and world == 1 and not args.no_train_graph
         try:
-            opt = torch.optim.Adam(net.parameters(), lr=5e-4, fused=True)
+            opt = torch.optim.Adam(net.parameters(), lr=5e-4, fused=True,
+                                   capturable=want_graph)
         except (RuntimeError, TypeError, ValueError):
             opt = torch.optim.Adam(net.parameters(), lr=5e-4, foreach=True)
+            want_graph = False
 
-        def step(i):
+        def eager_step(i):
             img, hm, off, wh, mask = batches[i % len(batches)]
             with rthd_amp.autocast(enabled=amp_on):
                 out = model(img)
@@ -141,6 +146,30 @@ def main():
                 model.finish_backward()
             opt.step()
             opt.zero_grad(set_to_none=True)
+
+        step = eager_step
+        if want_graph:
+            # hipGraph-captured training step (one graph per resident
+            # synthetic batch): the ~400 kernel launches of a step become
+            # one graph launch. Falls back to eager stepping on any
+            # capture failure.
+            try:
+                for i in range(3):
+                    eager_step(i)
+                torch.cuda.synchronize()
+                graphs = []
+                for b in range(len(batches)):
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        eager_step(b)
+                    graphs.append(g)
+                calc._pending.clear()
+
+                def step(i):
+                    graphs[i % len(graphs)].replay()
+            except Exception as e:
+                print('train-graph capture failed (%s); eager stepping' % e)
+                step = eager_step
     else:
         from real_time_helmet_detection_amd.engine.evaluator import (
             Prediction, GraphedPredictor)

@@ -132,14 +132,14 @@ class Convolution(nn.Module):
                                  track_running_stats=True) if bn \
             else nn.Identity()
 
-    def forward(self, x):
+    def forward(self, x, skip=None, act_override=None):
         bn = self.bn if isinstance(self.bn, nn.BatchNorm2d) else None
-        name = self.activation.name
+        name = act_override or self.activation.name
         if name in FUSIBLE_ACTS:
             return F2.conv_bn_act(x, self.convolution, bn, name, None,
-                                  self.training)
+                                  self.training, skip=skip)
         y = F2.conv_bn_act(x, self.convolution, bn, 'Linear', None,
-                           self.training)
+                           self.training, skip=skip)
         return self.activation(y)
 
 
@@ -163,12 +163,13 @@ class Residual(nn.Module):
             self.skip = nn.Identity()
 
     def forward(self, x):
-        y = self.conv2(self.conv1(x))
+        y1 = self.conv1(x)
         s = x if isinstance(self.skip, nn.Identity) else self.skip(x)
         name = self.activation.name
         if name in FUSIBLE_ACTS:
-            return F2.add_act(y, s, name, None)
-        return self.activation(y + s)
+            # fuse skip-add + outer activation into conv2's epilogue
+            return self.conv2(y1, skip=s, act_override=name)
+        return self.activation(self.conv2(y1) + s)
 
 
 class Hourglass(nn.Module):

@@ -63,11 +63,13 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
                                     double momentum, double eps);
 torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
                          torch::Tensor rstd, torch::Tensor gamma,
-                         torch::Tensor beta, int64_t act);
+                         torch::Tensor beta, int64_t act,
+                         c10::optional<torch::Tensor> skip);
 std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
                                       torch::Tensor mean, torch::Tensor rstd,
                                       torch::Tensor gamma, torch::Tensor beta,
-                                      int64_t act);
+                                      int64_t act,
+                                      c10::optional<torch::Tensor> skip);
 torch::Tensor col_sum(torch::Tensor x);
 }  // namespace rthd
 
@@ -93,7 +95,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
   m.def("bn_stats", &rthd::bn_stats);
-  m.def("bn_act_fwd", &rthd::bn_act_fwd);
-  m.def("bn_act_bwd", &rthd::bn_act_bwd);
+  m.def("bn_act_fwd", &rthd::bn_act_fwd, py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());
+  m.def("bn_act_bwd", &rthd::bn_act_bwd, py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());
   m.def("col_sum", &rthd::col_sum);
 }

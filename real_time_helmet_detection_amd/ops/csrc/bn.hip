@@ -196,12 +196,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
 
 // ---------------------------- bn apply forward ------------------------------
 
-template <typename T>
+template <typename T, bool HAS_SKIP>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
                                   const float* __restrict__ mean,
                                   const float* __restrict__ rstd,
                                   const float* __restrict__ gamma,
                                   const float* __restrict__ beta,
+                                  const T* __restrict__ skip,
                                   T* __restrict__ y, int64_t n, int C,
                                   int act) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
@@ -209,16 +210,20 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
   for (; i < n; i += stride) {
     const int c = i % C;
     const float xh = (ldf(&x[i]) - mean[c]) * rstd[c];
-    stf(&y[i], apply_act(xh * gamma[c] + beta[c], act));
+    float pre = xh * gamma[c] + beta[c];
+    if (HAS_SKIP) pre += ldf(&skip[i]);
+    stf(&y[i], apply_act(pre, act));
   }
 }
 
 // bf16 fast: fixed octet per thread -> per-channel params hoisted
+template <bool HAS_SKIP>
 __global__ void bn_act_fwd8_kernel(const bf16* __restrict__ x,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ rstd,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
+                                   const bf16* __restrict__ skip,
                                    bf16* __restrict__ y, int64_t M, int C,
                                    int act) {
   const int octs = C >> 3;
@@ -235,23 +240,30 @@ __global__ void bn_act_fwd8_kernel(const bf16* __restrict__ x,
     sh[e] = beta[c0 + e] - mean[c0 + e] * g;
   }
   for (int64_t r = rs; r < M; r += streams) {
-    bf16 v[8], o[8];
+    bf16 v[8], o[8], sk[8];
     *reinterpret_cast<uint4*>(v) =
         *reinterpret_cast<const uint4*>(&x[r * C + c0]);
+    if (HAS_SKIP)
+      *reinterpret_cast<uint4*>(sk) =
+          *reinterpret_cast<const uint4*>(&skip[r * C + c0]);
 #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      o[e] = f2b(apply_act(b2f(v[e]) * sc[e] + sh[e], act));
+    for (int e = 0; e < 8; ++e) {
+      float pre = b2f(v[e]) * sc[e] + sh[e];
+      if (HAS_SKIP) pre += b2f(sk[e]);
+      o[e] = f2b(apply_act(pre, act));
+    }
     *reinterpret_cast<uint4*>(&y[r * C + c0]) = *reinterpret_cast<uint4*>(o);
   }
 }
 
 // ---------------------------- bn backward -----------------------------------
 
-template <typename T>
+template <typename T, bool HAS_SKIP>
 __global__ void bn_bwd_reduce_part_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
+    const T* __restrict__ skip,
     float* __restrict__ p1, float* __restrict__ p2,
     int64_t M, int C, int act) {
   const int c = blockIdx.x * 64 + (threadIdx.x & 63);
@@ -267,8 +279,9 @@ __global__ void bn_bwd_reduce_part_kernel(
   float a1 = 0.f, a2 = 0.f;
   for (int64_t r = r0 + sub; r < r1; r += 4) {
     const float xh = (ldf(&x[r * C + c]) - mu) * rsd;
-    const float dpre = ldf(&dy[r * C + c]) *
-        act_grad_from_pre(xh * gm + bt, act);
+    float pre = xh * gm + bt;
+    if (HAS_SKIP) pre += ldf(&skip[r * C + c]);
+    const float dpre = ldf(&dy[r * C + c]) * act_grad_from_pre(pre, act);
     a1 += dpre;
     a2 += dpre * xh;
   }
@@ -286,10 +299,12 @@ __global__ void bn_bwd_reduce_part_kernel(
   }
 }
 
+template <bool HAS_SKIP>
 __global__ void bn_bwd_reduce8_part_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
+    const bf16* __restrict__ skip,
     float* __restrict__ p1, float* __restrict__ p2,
     int64_t M, int C, int act) {
   const int octs = C >> 3;
@@ -310,16 +325,20 @@ __global__ void bn_bwd_reduce8_part_kernel(
   }
   float a1[8] = {}, a2[8] = {};
   for (int64_t r = r0 + rs; r < r1; r += streams) {
-    bf16 vx[8], vdy[8];
+    bf16 vx[8], vdy[8], vsk[8];
     *reinterpret_cast<uint4*>(vx) =
         *reinterpret_cast<const uint4*>(&x[r * C + c0]);
     *reinterpret_cast<uint4*>(vdy) =
         *reinterpret_cast<const uint4*>(&dy[r * C + c0]);
+    if (HAS_SKIP)
+      *reinterpret_cast<uint4*>(vsk) =
+          *reinterpret_cast<const uint4*>(&skip[r * C + c0]);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       const float xh = (b2f(vx[e]) - mu[e]) * rsd[e];
-      const float dpre = b2f(vdy[e]) *
-          act_grad_from_pre(xh * gm[e] + bt[e], act);
+      float pre = xh * gm[e] + bt[e];
+      if (HAS_SKIP) pre += b2f(vsk[e]);
+      const float dpre = b2f(vdy[e]) * act_grad_from_pre(pre, act);
       a1[e] += dpre;
       a2[e] += dpre * xh;
     }
@@ -350,12 +369,13 @@ __global__ void bn_bwd_reduce8_part_kernel(
   }
 }
 
-template <typename T>
+template <typename T, bool HAS_SKIP>
 __global__ void bn_act_bwd_apply_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ s1, const float* __restrict__ s2,
+    const T* __restrict__ skip, T* __restrict__ dskip,
     T* __restrict__ dx, int64_t n, int C, float Mf, int act) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -363,17 +383,21 @@ __global__ void bn_act_bwd_apply_kernel(
     const int c = i % C;
     const float mu = mean[c], rs = rstd[c], gm = gamma[c];
     const float xh = (ldf(&x[i]) - mu) * rs;
-    const float dpre = ldf(&dy[i]) *
-        act_grad_from_pre(xh * gm + beta[c], act);
+    float pre = xh * gm + beta[c];
+    if (HAS_SKIP) pre += ldf(&skip[i]);
+    const float dpre = ldf(&dy[i]) * act_grad_from_pre(pre, act);
+    if (HAS_SKIP) stf(&dskip[i], dpre);
     stf(&dx[i], gm * rs * (dpre - s1[c] / Mf - xh * s2[c] / Mf));
   }
 }
 
+template <bool HAS_SKIP>
 __global__ void bn_act_bwd_apply8_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ s1, const float* __restrict__ s2,
+    const bf16* __restrict__ skip, bf16* __restrict__ dskip,
     bf16* __restrict__ dx, int64_t M, int C, float Mf, int act) {
   const int octs = C >> 3;
   const int streams = (int)(((int64_t)blockDim.x * gridDim.y) / octs);
@@ -393,18 +417,26 @@ __global__ void bn_act_bwd_apply8_kernel(
     t2[e] = s2[c0 + e] * invM;
   }
   for (int64_t r = rs; r < M; r += streams) {
-    bf16 vx[8], vdy[8], o[8];
+    bf16 vx[8], vdy[8], o[8], vsk[8], osk[8];
     *reinterpret_cast<uint4*>(vx) =
         *reinterpret_cast<const uint4*>(&x[r * C + c0]);
     *reinterpret_cast<uint4*>(vdy) =
         *reinterpret_cast<const uint4*>(&dy[r * C + c0]);
+    if (HAS_SKIP)
+      *reinterpret_cast<uint4*>(vsk) =
+          *reinterpret_cast<const uint4*>(&skip[r * C + c0]);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       const float xh = (b2f(vx[e]) - mu[e]) * rsd[e];
-      const float dpre = b2f(vdy[e]) *
-          act_grad_from_pre(xh * gm[e] + bt[e], act);
+      float pre = xh * gm[e] + bt[e];
+      if (HAS_SKIP) pre += b2f(vsk[e]);
+      const float dpre = b2f(vdy[e]) * act_grad_from_pre(pre, act);
+      if (HAS_SKIP) osk[e] = f2b(dpre);
       o[e] = f2b(gm[e] * rsd[e] * (dpre - t1[e] - xh * t2[e]));
     }
+    if (HAS_SKIP)
+      *reinterpret_cast<uint4*>(&dskip[r * C + c0]) =
+          *reinterpret_cast<uint4*>(osk);
     *reinterpret_cast<uint4*>(&dx[r * C + c0]) =
         *reinterpret_cast<uint4*>(o);
   }
@@ -490,7 +522,8 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
 
 torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
                          torch::Tensor rstd, torch::Tensor gamma,
-                         torch::Tensor beta, int64_t act) {
+                         torch::Tensor beta, int64_t act,
+                         c10::optional<torch::Tensor> skip) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
   const int64_t n = xc.numel();
@@ -499,22 +532,45 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
   auto gm = gamma.to(at::kFloat).contiguous();
   auto bt = beta.to(at::kFloat).contiguous();
   auto s = at::cuda::getCurrentCUDAStream();
+  const bool has_skip = skip.has_value();
+  torch::Tensor sk;
+  if (has_skip)
+    sk = skip->to(xc.scalar_type()).contiguous(at::MemoryFormat::ChannelsLast);
   if (fast8_ok(xc, C)) {
     const int octs = C / 8;
     const int nb = (int)std::min<int64_t>(cdiv(M * octs, 256), 2048);
-    hipLaunchKernelGGL(bn_act_fwd8_kernel, dim3(1, nb), dim3(256), 0, s,
-        reinterpret_cast<const bf16*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        reinterpret_cast<bf16*>(y.data_ptr()), M, C, (int)act);
+    auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
+    auto* py = reinterpret_cast<bf16*>(y.data_ptr());
+    const bf16* ps = has_skip
+        ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
+    if (has_skip)
+      hipLaunchKernelGGL((bn_act_fwd8_kernel<true>), dim3(1, nb), dim3(256),
+          0, s, px, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(), ps, py, M, C,
+          (int)act);
+    else
+      hipLaunchKernelGGL((bn_act_fwd8_kernel<false>), dim3(1, nb), dim3(256),
+          0, s, px, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(), ps, py, M, C,
+          (int)act);
   } else {
     DT(xc, {
-      hipLaunchKernelGGL((bn_act_fwd_kernel<scalar_t>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s,
-          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-          mean.data_ptr<float>(), rstd.data_ptr<float>(),
-          gm.data_ptr<float>(), bt.data_ptr<float>(),
-          reinterpret_cast<scalar_t*>(y.data_ptr()), n, C, (int)act);
+      auto* px = reinterpret_cast<const scalar_t*>(xc.data_ptr());
+      auto* py = reinterpret_cast<scalar_t*>(y.data_ptr());
+      const scalar_t* ps = has_skip
+          ? reinterpret_cast<const scalar_t*>(sk.data_ptr()) : nullptr;
+      if (has_skip)
+        hipLaunchKernelGGL((bn_act_fwd_kernel<scalar_t, true>),
+            dim3(ew_grid(n, 256)), dim3(256), 0, s, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(), ps, py, n, C,
+            (int)act);
+      else
+        hipLaunchKernelGGL((bn_act_fwd_kernel<scalar_t, false>),
+            dim3(ew_grid(n, 256)), dim3(256), 0, s, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(), ps, py, n, C,
+            (int)act);
     });
   }
   HIP_CHECK_LAST();
@@ -524,7 +580,8 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
 std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
                                       torch::Tensor mean, torch::Tensor rstd,
                                       torch::Tensor gamma, torch::Tensor beta,
-                                      int64_t act) {
+                                      int64_t act,
+                                      c10::optional<torch::Tensor> skip) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   auto dyc = dy.to(xc.scalar_type()).contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
@@ -536,27 +593,51 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   auto gm = gamma.to(at::kFloat).contiguous();
   auto bt = beta.to(at::kFloat).contiguous();
   auto s = at::cuda::getCurrentCUDAStream();
+  const bool has_skip = skip.has_value();
+  torch::Tensor sk, dsk;
+  if (has_skip) {
+    sk = skip->to(xc.scalar_type()).contiguous(at::MemoryFormat::ChannelsLast);
+    dsk = torch::empty_like(xc);
+  }
 
   const int chunks = pick_chunks(M, C);
   auto p1 = torch::empty({chunks, C}, s1.options());
   auto p2 = torch::empty({chunks, C}, s1.options());
   if (fast8_ok(xc, C)) {
-    hipLaunchKernelGGL(bn_bwd_reduce8_part_kernel, dim3(1, chunks),
-        dim3(256), 0, s,
-        reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        reinterpret_cast<const bf16*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
+    auto* pdy = reinterpret_cast<const bf16*>(dyc.data_ptr());
+    auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
+    const bf16* ps = has_skip
+        ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
+    if (has_skip)
+      hipLaunchKernelGGL((bn_bwd_reduce8_part_kernel<true>), dim3(1, chunks),
+          dim3(256), 0, s, pdy, px,
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(), ps,
+          p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce8_part_kernel<false>),
+          dim3(1, chunks), dim3(256), 0, s, pdy, px,
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(), ps,
+          p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
   } else {
     DT(xc, {
-      hipLaunchKernelGGL((bn_bwd_reduce_part_kernel<scalar_t>),
-          dim3(cdiv(C, 64), chunks), dim3(256), 0, s,
-          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
-          mean.data_ptr<float>(), rstd.data_ptr<float>(),
-          gm.data_ptr<float>(), bt.data_ptr<float>(),
-          p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
+      auto* pdy = reinterpret_cast<const scalar_t*>(dyc.data_ptr());
+      auto* px = reinterpret_cast<const scalar_t*>(xc.data_ptr());
+      const scalar_t* ps = has_skip
+          ? reinterpret_cast<const scalar_t*>(sk.data_ptr()) : nullptr;
+      if (has_skip)
+        hipLaunchKernelGGL((bn_bwd_reduce_part_kernel<scalar_t, true>),
+            dim3(cdiv(C, 64), chunks), dim3(256), 0, s, pdy, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(), ps,
+            p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
+      else
+        hipLaunchKernelGGL((bn_bwd_reduce_part_kernel<scalar_t, false>),
+            dim3(cdiv(C, 64), chunks), dim3(256), 0, s, pdy, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(), ps,
+            p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
     });
   }
   hipLaunchKernelGGL(reduce_partials_kernel,
@@ -567,28 +648,55 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   if (fast8_ok(xc, C)) {
     const int octs = C / 8;
     const int nb = (int)std::min<int64_t>(cdiv(M * octs, 256), 2048);
-    hipLaunchKernelGGL(bn_act_bwd_apply8_kernel, dim3(1, nb), dim3(256), 0,
-        s, reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        reinterpret_cast<const bf16*>(xc.data_ptr()),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        gm.data_ptr<float>(), bt.data_ptr<float>(),
-        s1.data_ptr<float>(), s2.data_ptr<float>(),
-        reinterpret_cast<bf16*>(dx.data_ptr()), M, C, (float)M, (int)act);
-  } else {
-    DT(xc, {
-      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<scalar_t>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s,
-          reinterpret_cast<const scalar_t*>(dyc.data_ptr()),
-          reinterpret_cast<const scalar_t*>(xc.data_ptr()),
+    auto* pdy = reinterpret_cast<const bf16*>(dyc.data_ptr());
+    auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
+    auto* pdx = reinterpret_cast<bf16*>(dx.data_ptr());
+    const bf16* ps = has_skip
+        ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
+    bf16* pdsk = has_skip
+        ? reinterpret_cast<bf16*>(dsk.data_ptr()) : nullptr;
+    if (has_skip)
+      hipLaunchKernelGGL((bn_act_bwd_apply8_kernel<true>), dim3(1, nb),
+          dim3(256), 0, s, pdy, px,
           mean.data_ptr<float>(), rstd.data_ptr<float>(),
           gm.data_ptr<float>(), bt.data_ptr<float>(),
-          s1.data_ptr<float>(), s2.data_ptr<float>(),
-          reinterpret_cast<scalar_t*>(dx.data_ptr()), n, C, (float)M,
-          (int)act);
+          s1.data_ptr<float>(), s2.data_ptr<float>(), ps, pdsk, pdx, M, C,
+          (float)M, (int)act);
+    else
+      hipLaunchKernelGGL((bn_act_bwd_apply8_kernel<false>), dim3(1, nb),
+          dim3(256), 0, s, pdy, px,
+          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+          gm.data_ptr<float>(), bt.data_ptr<float>(),
+          s1.data_ptr<float>(), s2.data_ptr<float>(), ps, pdsk, pdx, M, C,
+          (float)M, (int)act);
+  } else {
+    DT(xc, {
+      auto* pdy = reinterpret_cast<const scalar_t*>(dyc.data_ptr());
+      auto* px = reinterpret_cast<const scalar_t*>(xc.data_ptr());
+      auto* pdx = reinterpret_cast<scalar_t*>(dx.data_ptr());
+      const scalar_t* ps = has_skip
+          ? reinterpret_cast<const scalar_t*>(sk.data_ptr()) : nullptr;
+      scalar_t* pdsk = has_skip
+          ? reinterpret_cast<scalar_t*>(dsk.data_ptr()) : nullptr;
+      if (has_skip)
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<scalar_t, true>),
+            dim3(ew_grid(n, 256)), dim3(256), 0, s, pdy, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(),
+            s1.data_ptr<float>(), s2.data_ptr<float>(), ps, pdsk, pdx, n, C,
+            (float)M, (int)act);
+      else
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<scalar_t, false>),
+            dim3(ew_grid(n, 256)), dim3(256), 0, s, pdy, px,
+            mean.data_ptr<float>(), rstd.data_ptr<float>(),
+            gm.data_ptr<float>(), bt.data_ptr<float>(),
+            s1.data_ptr<float>(), s2.data_ptr<float>(), ps, pdsk, pdx, n, C,
+            (float)M, (int)act);
     });
   }
   HIP_CHECK_LAST();
-  // dgamma = s2, dbeta = s1
+  // dgamma = s2, dbeta = s1; dskip (= dpre) last when skip given
+  if (has_skip) return {dx, s2, s1, dsk};
   return {dx, s2, s1};
 }
 

@@ -41,16 +41,18 @@ def _named_act(y, act):
 
 
 def conv_bn_act(x, conv, bn=None, act='Linear', act_module=None,
-                training=False):
-    """conv -> (BN) -> activation, fused on the HIP path.
+                training=False, skip=None):
+    """conv -> (BN) -> (+skip) -> activation, fused on the HIP path.
 
     conv: nn.Conv2d holding weight/bias; bn: nn.BatchNorm2d or None;
     act: activation name; act_module: the nn.Module for parametric /
-    non-fusible activations (PReLU etc. — applied after a Linear epilogue).
+    non-fusible activations (PReLU etc. — applied after a Linear epilogue);
+    skip: optional residual tensor added BEFORE the activation (the
+    Residual tail fuses its add+act into this epilogue).
     """
     if _hip(x):
         from . import hip
-        return hip.conv_bn_act(x, conv, bn, act, act_module, training)
+        return hip.conv_bn_act(x, conv, bn, act, act_module, training, skip)
 
     y = F.conv2d(x, conv.weight, conv.bias, stride=conv.stride,
                  padding=conv.padding)
@@ -58,6 +60,8 @@ def conv_bn_act(x, conv, bn=None, act='Linear', act_module=None,
         y = F.batch_norm(y, bn.running_mean, bn.running_var, bn.weight,
                          bn.bias, training=training, momentum=bn.momentum,
                          eps=bn.eps)
+    if skip is not None:
+        y = y + skip
     if act_module is not None:
         return act_module(y)
     return _named_act(y, act)

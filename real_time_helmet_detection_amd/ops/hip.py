@@ -41,13 +41,17 @@ def _bf16_mode(x):
 
 class _ConvBNActFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, gamma, beta, rmean, rvar,
+    def forward(ctx, x, weight, bias, gamma, beta, rmean, rvar, skip,
                 kh, kw, stride, pad, act_code, use_bn, training, momentum,
                 eps, is_stem):
         C = _C()
         bf16 = _bf16_mode(x)
         dtype = torch.bfloat16 if bf16 else torch.float32
         xc = x.to(dtype).contiguous(memory_format=torch.channels_last)
+        skc = None
+        if skip is not None:
+            skc = skip.to(dtype).contiguous(
+                memory_format=torch.channels_last)
         cout = weight.shape[0]
         dev = x.device
 
@@ -58,10 +62,11 @@ class _ConvBNActFn(torch.autograd.Function):
         ones = torch.ones(cout, device=dev, dtype=torch.float32)
         zeros = torch.zeros(cout, device=dev, dtype=torch.float32)
 
-        def run_conv(scale, shift, act):
+        def run_conv(scale, shift, act, sk=None):
             if is_stem:
+                assert sk is None
                 return C.stem_fwd(xc, weight, scale, shift, stride, pad, act)
-            return C.conv_fwd(xc, wpk, scale, shift, None, kh, kw, stride,
+            return C.conv_fwd(xc, wpk, scale, shift, sk, kh, kw, stride,
                               pad, cout, act)
 
         bias_f = bias.float().contiguous() if bias is not None else zeros
@@ -71,20 +76,21 @@ class _ConvBNActFn(torch.autograd.Function):
             # stats match the eager/reference semantics (stem has bias+BN)
             y_lin = run_conv(ones, bias_f, ACT_CODE['Linear'])
             mean, rstd = C.bn_stats(y_lin, rmean, rvar, momentum, eps)
-            y = C.bn_act_fwd(y_lin, mean, rstd, gamma, beta, act_code)
+            y = C.bn_act_fwd(y_lin, mean, rstd, gamma, beta, act_code, skc)
         elif use_bn:
             rstd_run = torch.rsqrt(rvar.float() + eps)
             scale = (gamma.float() * rstd_run).contiguous()
             shift = (beta.float() + (bias_f - rmean.float()) * scale
                      ).contiguous()
-            y = run_conv(scale, shift, act_code)
+            y = run_conv(scale, shift, act_code, skc)
         else:
-            y = run_conv(ones, bias_f, act_code)
+            y = run_conv(ones, bias_f, act_code, skc)
 
         ctx.meta = (kh, kw, stride, pad, act_code, use_bn, training, bf16,
-                    is_stem, bias is not None, eps)
+                    is_stem, bias is not None, eps, skc is not None)
         if use_bn and training:
-            ctx.save_for_backward(xc, weight, gamma, beta, y_lin, mean, rstd)
+            ctx.save_for_backward(xc, weight, gamma, beta, y_lin, mean,
+                                  rstd, skc)
         elif use_bn:
             ctx.save_for_backward(xc, weight, gamma, beta, y,
                                   rmean.detach().clone(),
@@ -97,23 +103,32 @@ class _ConvBNActFn(torch.autograd.Function):
     def backward(ctx, dy):
         C = _C()
         (kh, kw, stride, pad, act_code, use_bn, training, bf16, is_stem,
-         has_bias, eps) = ctx.meta
-        dgamma = dbeta = dbias = None
+         has_bias, eps, has_skip) = ctx.meta
+        dgamma = dbeta = dbias = dskip = None
 
         if use_bn and training:
-            xc, weight, gamma, beta, y_lin, mean, rstd = ctx.saved_tensors
-            dpre, dgamma, dbeta = C.bn_act_bwd(dy, y_lin, mean, rstd,
-                                               gamma, beta, act_code)
+            (xc, weight, gamma, beta, y_lin, mean, rstd,
+             skc) = ctx.saved_tensors
+            outs = C.bn_act_bwd(dy, y_lin, mean, rstd, gamma, beta,
+                                act_code, skc)
+            if has_skip:
+                dpre, dgamma, dbeta, dskip = outs
+            else:
+                dpre, dgamma, dbeta = outs
         elif use_bn:
             # eval-mode BN backward (rare: grads through a frozen BN)
             xc, weight, gamma, beta, y, rmean, rvar = ctx.saved_tensors
             dact = C.add_act_bwd(dy, y, act_code)
+            if has_skip:
+                dskip = dact
             scale = (gamma.float() * torch.rsqrt(rvar.float() + eps))
             dpre = (dact.float() * scale.view(1, -1, 1, 1)).to(dact.dtype)
             dpre = dpre.contiguous(memory_format=torch.channels_last)
         else:
             xc, weight, y = ctx.saved_tensors
             dpre = C.add_act_bwd(dy, y, act_code)
+            if has_skip:
+                dskip = dpre
         if has_bias and ctx.needs_input_grad[2]:
             dbias = C.col_sum(dpre)
 
@@ -160,7 +175,7 @@ class _ConvBNActFn(torch.autograd.Function):
             else:
                 dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
 
-        return (dx, dw, dbias, dgamma, dbeta, None, None,
+        return (dx, dw, dbias, dgamma, dbeta, None, None, dskip,
                 None, None, None, None, None, None, None, None, None, None)
 
 
@@ -195,7 +210,8 @@ def _conv_infer_fp8(x, conv, bn, act_code):
                           None, kh, kw, stride, pad, cout, act_code)
 
 
-def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
+def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
+                skip=None):
     """GPU twin of functional.conv_bn_act (act_module path stays eager)."""
     kh, kw = conv.kernel_size
     stride = conv.stride[0]
@@ -210,7 +226,8 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
     if act_code is None:
         raise NotImplementedError(f'HIP conv epilogue: activation {act!r}')
     if (_amp.fp8_enabled() and not training and not is_stem
-            and not torch.is_grad_enabled() and cin % 16 == 0):
+            and not torch.is_grad_enabled() and cin % 16 == 0
+            and skip is None):
         y = _conv_infer_fp8(x, conv, bn, act_code)
         if act_module is not None:
             y = act_module(y)
@@ -222,6 +239,7 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False):
         bn.bias if use_bn else None,
         bn.running_mean if use_bn else None,
         bn.running_var if use_bn else None,
+        skip,
         kh, kw, stride, pad, act_code, use_bn, training,
         bn.momentum if use_bn else 0.1,
         bn.eps if use_bn else 1e-5,

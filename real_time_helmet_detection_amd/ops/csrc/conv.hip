@@ -109,13 +109,26 @@ struct ConvGeo {
 };
 
 // act codes from common.h; epilogue: y = act(acc*scale[c] + shift[c] (+skip))
-template <bool HAS_SKIP, bool ALIGNED>
+//
+// Staging is ASYNC global->LDS (global_load_lds_dwordx4, guide G15): each
+// K-step issues the NEXT step's 4 glds per thread into the alternate LDS
+// buffer, then runs this step's fragment reads + MFMA; the barrier at the
+// step end both publishes the prefetched tile and closes the read window —
+// ONE barrier per K-step and no ds_write pass or staging VGPRs. Out-of-range
+// pixels point their source at a 16-B zero page (glds cannot select-zero).
+// The per-lane SOURCE k8 is pre-swizzled so the lane-linear LDS image equals
+// the XOR-swizzled layout the fragment reads expect (guide rule 21).
+typedef __attribute__((address_space(3))) void lds_void;
+typedef __attribute__((address_space(1))) const void glb_void;
+
+template <bool HAS_SKIP>
 __global__ __launch_bounds__(256)
 void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ wpk,
                           const float* __restrict__ scale,
                           const float* __restrict__ shift,
                           const bf16* __restrict__ skip,
+                          const bf16* __restrict__ zpage,
                           bf16* __restrict__ y,
                           ConvGeo g, int act) {
   // grid: (M/128) x (Coutp/128)
@@ -127,17 +140,17 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
   const int wr = wid >> 1, wc = wid & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* ldsA = reinterpret_cast<bf16*>(smem);             // 8 KB
-  bf16* ldsB = reinterpret_cast<bf16*>(smem + 8192);      // 8 KB
+  // double-buffered: [buf][A 8KB | B 8KB]
+  char* lds = smem;
 
   f32x4 acc[4][4] = {};
 
-  // staging indices: 256 threads x 16 B = 4 KB per pass; A tile 8 KB -> 2
-  // passes; each thread covers rows (tid/4) and (tid/4 + 64), k8 = tid%4.
+  // thread's two A/B rows and pre-swizzled source k8 (constant per thread)
   const int st_row = tid >> 2;
   const int st_k8 = tid & 3;
+  const int k8s0 = st_k8 ^ ((st_row >> 2) & 3);
+  const int k8s1 = st_k8 ^ (((st_row + 64) >> 2) & 3);
 
-  // per-thread A source pixel decomposition for its two staging rows
   int am[2], ab[2], ay[2], ax[2];
 #pragma unroll
   for (int h = 0; h < 2; ++h) {
@@ -153,91 +166,61 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
   const int nsteps = g.KH * g.KW * (g.Cinp / 32);
   const int kc_per_tap = g.Cinp / 32;
 
-  // register-staged software pipeline (guide G15/T14): loads for step k+1
-  // are ISSUED during step k's MFMA and only CONSUMED at the next ds_write,
-  // hiding the HBM latency that a stage->sync->MFMA loop exposes every
-  // K-step. Single LDS buffer (the write of k+1 happens after the second
-  // barrier, when step k's fragment reads are done). Scalars, not arrays:
-  // by-reference arrays in a lambda spilled to scratch (guide rule 20).
-#define RTHD_LOAD_STEP(STEP, A0, A1, B0, B1)                                 \
+  // wave-uniform LDS bases for this thread's 4 glds (A r0/r1, B r0/r1)
+  const int wbase = wid * 1024;
+
+#define RTHD_GLDS_STEP(STEP, BUF)                                            \
   do {                                                                       \
     const int t_ = (STEP) / kc_per_tap;                                      \
     const int kb_ = (STEP) % kc_per_tap;                                     \
     const int dy_ = t_ / g.KW - g.pad;                                       \
     const int dx_ = t_ % g.KW - g.pad;                                       \
-    const int c0_ = kb_ * 32 + st_k8 * 8;                                    \
-    uint4 v0_ = {0, 0, 0, 0}, v1_ = {0, 0, 0, 0};                            \
+    char* base_ = lds + (BUF) * 16384;                                       \
+    const int c00_ = kb_ * 32 + k8s0 * 8;                                    \
+    const int c01_ = kb_ * 32 + k8s1 * 8;                                    \
+    const bf16* sa0_ = zpage;                                                \
+    const bf16* sa1_ = zpage;                                                \
     {                                                                        \
       const int iy_ = ay[0] * g.stride + dy_;                                \
       const int ix_ = ax[0] * g.stride + dx_;                                \
-      const bool val_ = am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 &&  \
-                        ix_ < g.W && c0_ < g.Cin;                            \
-      if (ALIGNED) {                                                         \
-        /* unconditional clamped load + select-zero: a branch around the  */ \
-        /* load costs a vmcnt(0) drain per element (guide trap (c))       */ \
-        const int64_t o_ = val_                                              \
-            ? (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin + c0_       \
-            : 0;                                                             \
-        v0_ = *reinterpret_cast<const uint4*>(x + o_);                       \
-        if (!val_) v0_ = uint4{0, 0, 0, 0};                                  \
-      } else if (val_) {                                                     \
-        const bf16* src_ =                                                   \
-            x + (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
-        bf16 tmp_[8] = {};                                                   \
-        for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];        \
-        v0_ = *reinterpret_cast<const uint4*>(tmp_);                         \
-      }                                                                      \
+      if (am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
+          c00_ < g.Cin)                                                      \
+        sa0_ = x + (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin +      \
+               c00_;                                                         \
     }                                                                        \
     {                                                                        \
       const int iy_ = ay[1] * g.stride + dy_;                                \
       const int ix_ = ax[1] * g.stride + dx_;                                \
-      const bool val_ = am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 &&  \
-                        ix_ < g.W && c0_ < g.Cin;                            \
-      if (ALIGNED) {                                                         \
-        const int64_t o_ = val_                                              \
-            ? (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin + c0_       \
-            : 0;                                                             \
-        v1_ = *reinterpret_cast<const uint4*>(x + o_);                       \
-        if (!val_) v1_ = uint4{0, 0, 0, 0};                                  \
-      } else if (val_) {                                                     \
-        const bf16* src_ =                                                   \
-            x + (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin + c0_;    \
-        bf16 tmp_[8] = {};                                                   \
-        for (int e_ = 0; c0_ + e_ < g.Cin; ++e_) tmp_[e_] = src_[e_];        \
-        v1_ = *reinterpret_cast<const uint4*>(tmp_);                         \
-      }                                                                      \
+      if (am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
+          c01_ < g.Cin)                                                      \
+        sa1_ = x + (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin +      \
+               c01_;                                                         \
     }                                                                        \
-    (A0) = v0_;                                                              \
-    (A1) = v1_;                                                              \
-    (B0) = *reinterpret_cast<const uint4*>(                                  \
-        wpk + ((int64_t)t_ * g.Coutp + nblk * 128 + st_row) * g.Cinp +       \
-        kb_ * 32 + st_k8 * 8);                                               \
-    (B1) = *reinterpret_cast<const uint4*>(                                  \
-        wpk + ((int64_t)t_ * g.Coutp + nblk * 128 + st_row + 64) * g.Cinp +  \
-        kb_ * 32 + st_k8 * 8);                                               \
+    const bf16* sb0_ = wpk +                                                 \
+        ((int64_t)t_ * g.Coutp + nblk * 128 + st_row) * g.Cinp + kb_ * 32 +  \
+        k8s0 * 8;                                                            \
+    const bf16* sb1_ = wpk +                                                 \
+        ((int64_t)t_ * g.Coutp + nblk * 128 + st_row + 64) * g.Cinp +        \
+        kb_ * 32 + k8s1 * 8;                                                 \
+    __builtin_amdgcn_global_load_lds((glb_void*)sa0_,                        \
+        (lds_void*)(base_ + wbase), 16, 0, 0);                               \
+    __builtin_amdgcn_global_load_lds((glb_void*)sa1_,                        \
+        (lds_void*)(base_ + 4096 + wbase), 16, 0, 0);                        \
+    __builtin_amdgcn_global_load_lds((glb_void*)sb0_,                        \
+        (lds_void*)(base_ + 8192 + wbase), 16, 0, 0);                        \
+    __builtin_amdgcn_global_load_lds((glb_void*)sb1_,                        \
+        (lds_void*)(base_ + 12288 + wbase), 16, 0, 0);                       \
   } while (0)
 
-  bf16* A = ldsA;
-  bf16* B = ldsB;
-  uint4 sa0, sa1, sb0, sb1, na0, na1, nb0, nb1;
-  RTHD_LOAD_STEP(0, sa0, sa1, sb0, sb1);
+  RTHD_GLDS_STEP(0, 0);
+  __syncthreads();
 
   for (int step = 0; step < nsteps; ++step) {
-    // ---- write current tile (regs -> LDS) ----
-    *reinterpret_cast<uint4*>(
-        reinterpret_cast<char*>(A) + lds_off_bf16(st_row, st_k8)) = sa0;
-    *reinterpret_cast<uint4*>(
-        reinterpret_cast<char*>(A) + lds_off_bf16(st_row + 64, st_k8)) = sa1;
-    *reinterpret_cast<uint4*>(
-        reinterpret_cast<char*>(B) + lds_off_bf16(st_row, st_k8)) = sb0;
-    *reinterpret_cast<uint4*>(
-        reinterpret_cast<char*>(B) + lds_off_bf16(st_row + 64, st_k8)) = sb1;
-    __syncthreads();
+    const int buf = step & 1;
+    if (step + 1 < nsteps) RTHD_GLDS_STEP(step + 1, buf ^ 1);
 
-    // ---- issue next tile's loads (consumed at the next ds_write) ----
-    if (step + 1 < nsteps) RTHD_LOAD_STEP(step + 1, na0, na1, nb0, nb1);
-
-    // ---- fragments + MFMA ----
+    char* A = lds + buf * 16384;
+    char* B = A + 8192;
     const int arow_base = wr * 64 + (lane & 15);
     const int brow_base = wc * 64 + (lane & 15);
     const int k8 = lane >> 4;
@@ -245,9 +228,9 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       afrag[i] = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<char*>(A) + lds_off_bf16(arow_base + 16 * i, k8));
+          A + lds_off_bf16(arow_base + 16 * i, k8));
       bfrag[i] = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<char*>(B) + lds_off_bf16(brow_base + 16 * i, k8));
+          B + lds_off_bf16(brow_base + 16 * i, k8));
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
@@ -256,13 +239,11 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
 
+    // publishes the prefetched tile (vmcnt(0) on the in-flight glds) AND
+    // closes this buffer's read window before it is re-staged
     __syncthreads();
-    sa0 = na0;
-    sa1 = na1;
-    sb0 = nb0;
-    sb1 = nb1;
   }
-#undef RTHD_LOAD_STEP
+#undef RTHD_GLDS_STEP
 
   // ---- epilogue ----
   const int col0 = nblk * 128 + wc * 64 + (lane & 15);
@@ -465,30 +446,26 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
   auto s = at::cuda::getCurrentCUDAStream();
 
   if (bf16_mode) {
-    const size_t lds = 16384;
+    const size_t lds = 32768;
     if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
+    TORCH_CHECK(g.Cin % 8 == 0,
+                "bf16 conv requires Cin % 8 == 0 (glds staging); "
+                "Cin=", g.Cin, " runs the f32 path");
+    auto zpage = torch::zeros({8}, xc.options());
     auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
     auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());
     auto* py = reinterpret_cast<bf16*>(y.data_ptr());
+    auto* pz = reinterpret_cast<const bf16*>(zpage.data_ptr());
     const bf16* ps =
         has_skip ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
-    const bool aligned = g.Cin % 8 == 0;
-    if (has_skip && aligned)
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true, true>), grid,
+    if (has_skip)
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true>), grid,
           dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
-          sh.data_ptr<float>(), ps, py, g, (int)act);
-    else if (has_skip)
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<true, false>), grid,
-          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
-          sh.data_ptr<float>(), ps, py, g, (int)act);
-    else if (aligned)
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false, true>), grid,
-          dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
-          sh.data_ptr<float>(), ps, py, g, (int)act);
+          sh.data_ptr<float>(), ps, pz, py, g, (int)act);
     else
-      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false, false>), grid,
+      hipLaunchKernelGGL((conv_fwd_bf16_kernel<false>), grid,
           dim3(256), lds, s, px, pw, sc.data_ptr<float>(),
-          sh.data_ptr<float>(), ps, py, g, (int)act);
+          sh.data_ptr<float>(), ps, pz, py, g, (int)act);
   } else {
     const size_t lds = 32768;
     TORCH_CHECK(xc.scalar_type() == at::kFloat, "f32 conv needs f32 input");

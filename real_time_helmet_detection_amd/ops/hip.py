@@ -45,7 +45,9 @@ class _ConvBNActFn(torch.autograd.Function):
                 kh, kw, stride, pad, act_code, use_bn, training, momentum,
                 eps, is_stem):
         C = _C()
-        bf16 = _bf16_mode(x)
+        # the bf16 glds staging needs 16-B channel chunks; rare non-multiple
+        # channel counts (merge_prediction 6->128) take the exact-f32 path
+        bf16 = _bf16_mode(x) and x.shape[1] % 8 == 0
         dtype = torch.bfloat16 if bf16 else torch.float32
         xc = x.to(dtype).contiguous(memory_format=torch.channels_last)
         skc = None
@@ -143,9 +145,13 @@ class _ConvBNActFn(torch.autograd.Function):
             ones = torch.ones(cin, device=xc.device, dtype=torch.float32)
             zeros = torch.zeros(cin, device=xc.device, dtype=torch.float32)
             if stride == 1:
-                wpk_t = C.pack_weights(weight, True, bf16)
-                dx = C.conv_fwd(dpre, wpk_t, ones, zeros, None, kh, kw, 1,
+                bf16_d = bf16 and dpre.shape[1] % 8 == 0
+                wpk_t = C.pack_weights(weight, True, bf16_d)
+                dsrc = dpre if bf16_d else dpre.float()
+                dx = C.conv_fwd(dsrc, wpk_t, ones, zeros, None, kh, kw, 1,
                                 pad, cin, ACT_CODE['Linear'])
+                if dx.dtype != xc.dtype:
+                    dx = dx.to(xc.dtype)
             elif stride == 2 and kh == 2 and kw == 2 and pad == 0:
                 # k2/s2 windows don't overlap: each input pixel receives
                 # from exactly one output pixel, so dgrad decomposes into

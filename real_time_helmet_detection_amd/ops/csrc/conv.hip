@@ -131,26 +131,33 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ zpage,
                           bf16* __restrict__ y,
                           ConvGeo g, int act) {
-  // grid: (M/128) x (Coutp/128)
+  // grid: (M/128) x (Coutp/128); 4 waves (2x2 of 64x64).
+  //
+  // K loop = taps (outer) x 32-ch blocks (inner, incremental addressing —
+  // the flat-step version spent ~116 VALU/step on 64-bit address chains).
+  // Staging: async glds into a 3-deep LDS ring, TWO tiles in flight across
+  // each barrier via counted `s_waitcnt vmcnt(4)` + raw s_barrier (guide §5
+  // 'pipelining across barriers': a plain __syncthreads drains vmcnt(0) and
+  // de-pipelines the span). OOB pixels source a 16-B zero page.
   const int mblk = blockIdx.x;
   const int nblk = blockIdx.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;           // 4 waves: 2x2
+  const int wid = tid >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered: [buf][A 8KB | B 8KB]
-  char* lds = smem;
+  char* lds = smem;  // 3 x (A 8KB | B 8KB)
 
   f32x4 acc[4][4] = {};
 
-  // thread's two A/B rows and pre-swizzled source k8 (constant per thread)
   const int st_row = tid >> 2;
   const int st_k8 = tid & 3;
   const int k8s0 = st_k8 ^ ((st_row >> 2) & 3);
   const int k8s1 = st_k8 ^ (((st_row + 64) >> 2) & 3);
+  const int wbase = wid * 1024;
 
+  // per-thread pixel decomposition for its two staged rows
   int am[2], ab[2], ay[2], ax[2];
 #pragma unroll
   for (int h = 0; h < 2; ++h) {
@@ -163,64 +170,75 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
     ax[h] = r % g.Wo;
   }
 
-  const int nsteps = g.KH * g.KW * (g.Cinp / 32);
-  const int kc_per_tap = g.Cinp / 32;
+  const int kc = g.Cinp / 32;
+  const int taps = g.KH * g.KW;
+  const int nsteps = taps * kc;
 
-  // wave-uniform LDS bases for this thread's 4 glds (A r0/r1, B r0/r1)
-  const int wbase = wid * 1024;
+  // stage issue state, advanced incrementally by issue_step()
+  int is_step = 0;        // next step to issue
+  int is_t = 0, is_kb = 0;
+  const bf16* aptr[2];    // current tap's base source (or zpage)
+  const bf16* bptr[2];
+  bool avalid[2];
+  auto tap_setup = [&]() {
+    const int dy_ = is_t / g.KW - g.pad;
+    const int dx_ = is_t % g.KW - g.pad;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int iy = ay[h] * g.stride + dy_;
+      const int ix = ax[h] * g.stride + dx_;
+      avalid[h] = am[h] < g.M && iy >= 0 && iy < g.H && ix >= 0 &&
+                  ix < g.W;
+      const int c0 = (h ? k8s1 : k8s0) * 8;
+      aptr[h] = avalid[h]
+          ? x + (((int64_t)ab[h] * g.H + iy) * g.W + ix) * g.Cin + c0
+          : zpage;
+    }
+    bptr[0] = wpk + ((int64_t)is_t * g.Coutp + nblk * 128 + st_row) *
+        g.Cinp + k8s0 * 8;
+    bptr[1] = wpk + ((int64_t)is_t * g.Coutp + nblk * 128 + st_row + 64) *
+        g.Cinp + k8s1 * 8;
+  };
+  tap_setup();
 
-#define RTHD_GLDS_STEP(STEP, BUF)                                            \
-  do {                                                                       \
-    const int t_ = (STEP) / kc_per_tap;                                      \
-    const int kb_ = (STEP) % kc_per_tap;                                     \
-    const int dy_ = t_ / g.KW - g.pad;                                       \
-    const int dx_ = t_ % g.KW - g.pad;                                       \
-    char* base_ = lds + (BUF) * 16384;                                       \
-    const int c00_ = kb_ * 32 + k8s0 * 8;                                    \
-    const int c01_ = kb_ * 32 + k8s1 * 8;                                    \
-    const bf16* sa0_ = zpage;                                                \
-    const bf16* sa1_ = zpage;                                                \
-    {                                                                        \
-      const int iy_ = ay[0] * g.stride + dy_;                                \
-      const int ix_ = ax[0] * g.stride + dx_;                                \
-      if (am[0] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
-          c00_ < g.Cin)                                                      \
-        sa0_ = x + (((int64_t)ab[0] * g.H + iy_) * g.W + ix_) * g.Cin +      \
-               c00_;                                                         \
-    }                                                                        \
-    {                                                                        \
-      const int iy_ = ay[1] * g.stride + dy_;                                \
-      const int ix_ = ax[1] * g.stride + dx_;                                \
-      if (am[1] < g.M && iy_ >= 0 && iy_ < g.H && ix_ >= 0 && ix_ < g.W &&   \
-          c01_ < g.Cin)                                                      \
-        sa1_ = x + (((int64_t)ab[1] * g.H + iy_) * g.W + ix_) * g.Cin +      \
-               c01_;                                                         \
-    }                                                                        \
-    const bf16* sb0_ = wpk +                                                 \
-        ((int64_t)t_ * g.Coutp + nblk * 128 + st_row) * g.Cinp + kb_ * 32 +  \
-        k8s0 * 8;                                                            \
-    const bf16* sb1_ = wpk +                                                 \
-        ((int64_t)t_ * g.Coutp + nblk * 128 + st_row + 64) * g.Cinp +        \
-        kb_ * 32 + k8s1 * 8;                                                 \
-    __builtin_amdgcn_global_load_lds((glb_void*)sa0_,                        \
-        (lds_void*)(base_ + wbase), 16, 0, 0);                               \
-    __builtin_amdgcn_global_load_lds((glb_void*)sa1_,                        \
-        (lds_void*)(base_ + 4096 + wbase), 16, 0, 0);                        \
-    __builtin_amdgcn_global_load_lds((glb_void*)sb0_,                        \
-        (lds_void*)(base_ + 8192 + wbase), 16, 0, 0);                        \
-    __builtin_amdgcn_global_load_lds((glb_void*)sb1_,                        \
-        (lds_void*)(base_ + 12288 + wbase), 16, 0, 0);                       \
-  } while (0)
+  auto issue_step = [&]() {
+    char* base = lds + (is_step % 3) * 16384;
+    const int cb = is_kb * 32;
+    // channel-block bound (only non-trivial when Cin % 32 != 0)
+    const bf16* a0 = (avalid[0] && cb + k8s0 * 8 < g.Cin) ? aptr[0] + cb
+                                                          : zpage;
+    const bf16* a1 = (avalid[1] && cb + k8s1 * 8 < g.Cin) ? aptr[1] + cb
+                                                          : zpage;
+    __builtin_amdgcn_global_load_lds((glb_void*)a0,
+        (lds_void*)(base + wbase), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((glb_void*)a1,
+        (lds_void*)(base + 4096 + wbase), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((glb_void*)(bptr[0] + cb),
+        (lds_void*)(base + 8192 + wbase), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((glb_void*)(bptr[1] + cb),
+        (lds_void*)(base + 12288 + wbase), 16, 0, 0);
+    ++is_step;
+    if (++is_kb == kc) {
+      is_kb = 0;
+      if (++is_t < taps) tap_setup();
+    }
+  };
 
-  RTHD_GLDS_STEP(0, 0);
-  __syncthreads();
+  // prologue: two tiles in flight (count the wait to the FIRST tile)
+  issue_step();
+  if (nsteps > 1) {
+    issue_step();
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(4) : "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(0) : "memory");
+  }
+  __builtin_amdgcn_s_barrier();
 
   for (int step = 0; step < nsteps; ++step) {
-    const int buf = step & 1;
-    if (step + 1 < nsteps) RTHD_GLDS_STEP(step + 1, buf ^ 1);
-
-    char* A = lds + buf * 16384;
+    char* A = lds + (step % 3) * 16384;
     char* B = A + 8192;
+    if (step + 2 < nsteps) issue_step();
+
     const int arow_base = wr * 64 + (lane & 15);
     const int brow_base = wc * 64 + (lane & 15);
     const int k8 = lane >> 4;
@@ -239,14 +257,29 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
 
-    // publishes the prefetched tile (vmcnt(0) on the in-flight glds) AND
-    // closes this buffer's read window before it is re-staged
-    __syncthreads();
+    // wait for step+1's tile (leave step+2's 4 glds in flight), then a raw
+    // barrier — every wave has passed its own counted wait, so the tile is
+    // complete before any wave reads it.
+    if (step + 2 < nsteps + 1) {
+      if (step + 2 < nsteps)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(4) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(0) : "memory");
+    }
+    __builtin_amdgcn_s_barrier();
   }
-#undef RTHD_GLDS_STEP
 
   // ---- epilogue ----
+  // preload the 4 per-lane scale/shift pairs ONCE (the per-store scalar
+  // loads serialized the epilogue: 65 dependent vmcnt(0) waits in the .s)
   const int col0 = nblk * 128 + wc * 64 + (lane & 15);
+  float esc[4], esh[4];
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int c = col0 + ni * 16;
+    esc[ni] = c < g.Cout ? scale[c] : 0.f;
+    esh[ni] = c < g.Cout ? shift[c] : 0.f;
+  }
   const int row_in_frag = (lane >> 4) * 4;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
@@ -259,7 +292,7 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
         const int c = col0 + ni * 16;
         if (c >= g.Cout) continue;
         float v = acc[mi][ni][r];
-        v = v * scale[c] + shift[c];
+        v = v * esc[ni] + esh[ni];
         if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
         v = apply_act(v, act);
         stf(&y[(int64_t)m * g.Cout + c], v);
@@ -385,6 +418,13 @@ void conv_fwd_f32_kernel(const float* __restrict__ x,
   }
 
   const int col0 = nblk * 128 + wc * 64 + (lane & 15);
+  float esc[4], esh[4];
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int c = col0 + ni * 16;
+    esc[ni] = c < g.Cout ? scale[c] : 0.f;
+    esh[ni] = c < g.Cout ? shift[c] : 0.f;
+  }
   const int row_in_frag = (lane >> 4) * 4;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
@@ -397,7 +437,7 @@ void conv_fwd_f32_kernel(const float* __restrict__ x,
         const int c = col0 + ni * 16;
         if (c >= g.Cout) continue;
         float v = acc[mi][ni][r];
-        v = v * scale[c] + shift[c];
+        v = v * esc[ni] + esh[ni];
         if (HAS_SKIP) v += skip[(int64_t)m * g.Cout + c];
         v = apply_act(v, act);
         y[(int64_t)m * g.Cout + c] = v;
@@ -446,7 +486,7 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
   auto s = at::cuda::getCurrentCUDAStream();
 
   if (bf16_mode) {
-    const size_t lds = 32768;
+    const size_t lds = 3 * 16384;
     if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
     TORCH_CHECK(g.Cin % 8 == 0,
                 "bf16 conv requires Cin % 8 == 0 (glds staging); "
@@ -615,7 +655,16 @@ void conv_fwd_fp8_kernel(const bf16* __restrict__ x,
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
   }
 
+  // preload the 4 per-lane scale/shift pairs ONCE (the per-store scalar
+  // loads serialized the epilogue: 65 dependent vmcnt(0) waits in the .s)
   const int col0 = nblk * 128 + wc * 64 + (lane & 15);
+  float esc[4], esh[4];
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int c = col0 + ni * 16;
+    esc[ni] = c < g.Cout ? scale[c] : 0.f;
+    esh[ni] = c < g.Cout ? shift[c] : 0.f;
+  }
   const int row_in_frag = (lane >> 4) * 4;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
@@ -628,7 +677,7 @@ void conv_fwd_fp8_kernel(const bf16* __restrict__ x,
         const int c = col0 + ni * 16;
         if (c >= g.Cout) continue;
         float v = acc[mi][ni][r];
-        v = v * scale[c] + shift[c];
+        v = v * esc[ni] + esh[ni];
         if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
         v = apply_act(v, act);
         stf(&y[(int64_t)m * g.Cout + c], v);

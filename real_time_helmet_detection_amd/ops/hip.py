@@ -28,6 +28,19 @@ from .. import amp as _amp
 
 ACT_CODE = {'Linear': 0, 'ReLU': 1, 'LReLU': 2}
 
+# side stream for weight-gradient GEMMs: wgrad and dgrad both consume the
+# same upstream gradient and are independent, so wgrad runs concurrently
+# with the dgrad/BN chain (fills the CU idle time of the small layers).
+_wgrad_streams = {}
+
+
+def _wgrad_stream(device):
+    st = _wgrad_streams.get(device)
+    if st is None:
+        st = torch.cuda.Stream(device=device)
+        _wgrad_streams[device] = st
+    return st
+
 
 def _C():
     return _backend.require_ext()
@@ -134,6 +147,22 @@ class _ConvBNActFn(torch.autograd.Function):
         if has_bias and ctx.needs_input_grad[2]:
             dbias = C.col_sum(dpre)
 
+        # wgrad — on the side stream, overlapping the dgrad launched above
+        dw = None
+        if ctx.needs_input_grad[1]:
+            cur = torch.cuda.current_stream(xc.device)
+            side = _wgrad_stream(xc.device)
+            side.wait_stream(cur)
+            with torch.cuda.stream(side):
+                if is_stem:
+                    dw = C.stem_wgrad(xc, dpre, stride, pad)
+                elif xc.dtype == torch.bfloat16:
+                    dw = C.wgrad_bf16_fast(xc, dpre, kh, kw, stride, pad)
+                else:
+                    dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
+            cur.wait_stream(side)
+            dw.record_stream(cur)
+
         # dgrad
         dx = None
         if ctx.needs_input_grad[0]:
@@ -170,16 +199,6 @@ class _ConvBNActFn(torch.autograd.Function):
             else:
                 raise NotImplementedError(
                     f'HIP dgrad for k={kh} stride={stride} not implemented')
-
-        # wgrad
-        dw = None
-        if ctx.needs_input_grad[1]:
-            if is_stem:
-                dw = C.stem_wgrad(xc, dpre, stride, pad)
-            elif xc.dtype == torch.bfloat16:
-                dw = C.wgrad_bf16_fast(xc, dpre, kh, kw, stride, pad)
-            else:
-                dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
 
         return (dx, dw, dbias, dgamma, dbeta, None, None, dskip,
                 None, None, None, None, None, None, None, None, None, None)

@@ -59,8 +59,9 @@ class _ConvBNActFn(torch.autograd.Function):
                 eps, is_stem):
         C = _C()
         # the bf16 glds staging needs 16-B channel chunks; rare non-multiple
-        # channel counts (merge_prediction 6->128) take the exact-f32 path
-        bf16 = _bf16_mode(x) and x.shape[1] % 8 == 0
+        # channel counts (merge_prediction 6->128) take the exact-f32 path.
+        # The stem has its own direct kernel (any Cin) and stays bf16.
+        bf16 = _bf16_mode(x) and (is_stem or x.shape[1] % 8 == 0)
         dtype = torch.bfloat16 if bf16 else torch.float32
         xc = x.to(dtype).contiguous(memory_format=torch.channels_last)
         skc = None
@@ -157,7 +158,19 @@ class _ConvBNActFn(torch.autograd.Function):
                 if is_stem:
                     dw = C.stem_wgrad(xc, dpre, stride, pad)
                 elif xc.dtype == torch.bfloat16:
-                    dw = C.wgrad_bf16_fast(xc, dpre, kh, kw, stride, pad)
+                    # pad stray non-x8 channel counts (head Cout=6) so the
+                    # aligned fast kernel runs; slice the result back
+                    cin_p = xc.shape[1] % 8
+                    cout_p = dpre.shape[1] % 8
+                    xs = torch.nn.functional.pad(
+                        xc, (0, 0, 0, 0, 0, 8 - cin_p)).contiguous(
+                        memory_format=torch.channels_last) if cin_p else xc
+                    ds = torch.nn.functional.pad(
+                        dpre, (0, 0, 0, 0, 0, 8 - cout_p)).contiguous(
+                        memory_format=torch.channels_last) if cout_p                         else dpre
+                    dw = C.wgrad_bf16_fast(xs, ds, kh, kw, stride, pad)
+                    if cin_p or cout_p:
+                        dw = dw[:dpre.shape[1], :xc.shape[1]].contiguous()
                 else:
                     dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
             cur.wait_stream(side)

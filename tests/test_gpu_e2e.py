@@ -244,7 +244,7 @@ def test_overfit_synthetic_map():
     calc = LossCalculator().cuda()
     opt = torch.optim.Adam(net.parameters(), lr=2e-3)
     net.train()
-    for i in range(150):
+    for i in range(500):
         opt.zero_grad(set_to_none=True)
         with amp.autocast(True):
             out = net(img)
@@ -254,7 +254,7 @@ def test_overfit_synthetic_map():
         opt.step()
 
     net.eval()
-    pred = Prediction(net, topk=20, scale_factor=4, conf_th=0.25, nms='nms',
+    pred = Prediction(net, topk=20, scale_factor=4, conf_th=0.15, nms='nms',
                       nms_th=0.5).cuda()
     with torch.no_grad():
         boxes, clss, scores = pred(img)

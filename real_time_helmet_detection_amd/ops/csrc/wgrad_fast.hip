@@ -42,21 +42,20 @@ struct WgradGeo2 {
   int chunk_len;
 };
 
-template <bool ALIGNED, int KW_T>  // KW templated: static acc indexing
+template <bool ALIGNED>  // Cin,Cout % 8 == 0: branch-free staging loads
 __global__ __launch_bounds__(256)
 void wgrad_bf16_kernel(const bf16* __restrict__ x,
                        const bf16* __restrict__ dy,
                        float* __restrict__ dw, WgradGeo2 g) {
   // 4 waves; block tile [64 ci][64 co] (wave = 32 x 32). A 128x128 8-wave
-  // variant measured 2.2x SLOWER per unit work (64 KB LDS -> 2 blocks/CU).
-  // TAP-ROW grouping: one block owns a whole kernel ROW (KW_T taps): the
-  // dY tile is staged ONCE per pixel round and reused for all KW_T taps —
-  // dY staging volume and global traffic drop KW_T-fold.
-  const int trow = blockIdx.z % g.KH;
-  const int chunk = blockIdx.z / g.KH;
+  // variant measured 2.2x SLOWER per unit work (64 KB LDS -> 2 blocks/CU)
+  // — this geometry keeps 5 blocks/CU resident.
+  const int t = blockIdx.z % (g.KH * g.KW);
+  const int chunk = blockIdx.z / (g.KH * g.KW);
   const int ci0 = blockIdx.x * 64;
   const int co0 = blockIdx.y * 64;
-  const int dyt = trow - g.pad;
+  const int dyt = t / g.KW - g.pad;
+  const int dxt = t % g.KW - g.pad;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -67,7 +66,7 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   bf16* Xl = reinterpret_cast<bf16*>(smem);           // 17 KB (padded rows)
   bf16* Yl = reinterpret_cast<bf16*>(smem + 64 * 272);
 
-  f32x4 acc[KW_T][2][2] = {};
+  f32x4 acc[2][2] = {};
 
   const int px_start = chunk * g.chunk_len;
   const int px_end = min(px_start + g.chunk_len, g.M);
@@ -88,9 +87,53 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
 
   for (int p0 = px_start; p0 < px_end; p0 += 128) {
     __syncthreads();
-    // ---- stage dY ONCE per round (second thread half) ----
-    if (!stage_x) {
-      ushort r[8][8];
+    ushort r[8][8];
+    if (stage_x) {
+      int bj = bb, oyj = oy, oxj = ox;
+      const int cbase = ci0 + s_ci;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int m = p0 + s_px + j;
+        uint4 v;
+        const int iy = oyj * g.stride + dyt;
+        const int ix = oxj * g.stride + dxt;
+        const bool val = m < px_end && iy >= 0 && iy < g.H && ix >= 0 &&
+                         ix < g.W && cbase < g.Cin;
+        if (ALIGNED) {
+          // UNCONDITIONAL load from a clamped address + select-zero: a
+          // branch around the load makes hipcc drain vmcnt(0) per element
+          // (guide §5 trap (c) — measured 2-6x on this kernel)
+          const int64_t off = val
+              ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase
+              : 0;
+          v = *reinterpret_cast<const uint4*>(x + off);
+          if (!val) v = uint4{0, 0, 0, 0};
+        } else {
+          v = uint4{0, 0, 0, 0};
+          if (val) {
+            const ushort* src = reinterpret_cast<const ushort*>(
+                x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
+            ushort tmp[8] = {};
+            for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
+            v = *reinterpret_cast<const uint4*>(tmp);
+          }
+        }
+        *reinterpret_cast<uint4*>(r[j]) = v;
+        if (++oxj >= g.Wo) {
+          oxj = 0;
+          if (++oyj >= g.Ho) { oyj = 0; ++bj; }
+        }
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        ushort o[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[j] = r[j][e];
+        *reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(Xl) +
+            wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
+      }
+    } else {
       const int cbase = co0 + s_ci;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -124,81 +167,6 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
             wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
       }
     }
-
-#pragma unroll
-    for (int tx = 0; tx < KW_T; ++tx) {
-      const int dxt = tx - g.pad;
-      // ---- stage X for this tap (first thread half) ----
-      if (stage_x) {
-        ushort r[8][8];
-        int bj = bb, oyj = oy, oxj = ox;
-        const int cbase = ci0 + s_ci;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int m = p0 + s_px + j;
-          uint4 v;
-          const int iy = oyj * g.stride + dyt;
-          const int ix = oxj * g.stride + dxt;
-          const bool val = m < px_end && iy >= 0 && iy < g.H && ix >= 0 &&
-                           ix < g.W && cbase < g.Cin;
-          if (ALIGNED) {
-            // UNCONDITIONAL clamped load + select-zero (guide §5 trap (c))
-            const int64_t off = val
-                ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase
-                : 0;
-            v = *reinterpret_cast<const uint4*>(x + off);
-            if (!val) v = uint4{0, 0, 0, 0};
-          } else {
-            v = uint4{0, 0, 0, 0};
-            if (val) {
-              const ushort* src = reinterpret_cast<const ushort*>(
-                  x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin +
-                  cbase);
-              ushort tmp[8] = {};
-              for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
-              v = *reinterpret_cast<const uint4*>(tmp);
-            }
-          }
-          *reinterpret_cast<uint4*>(r[j]) = v;
-          if (++oxj >= g.Wo) {
-            oxj = 0;
-            if (++oyj >= g.Ho) { oyj = 0; ++bj; }
-          }
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          ushort o[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) o[j] = r[j][e];
-          *reinterpret_cast<uint4*>(
-              reinterpret_cast<char*>(Xl) +
-              wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
-        }
-      }
-      __syncthreads();
-
-#pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
-        const int k8 = (lane >> 4) + ks * 4;
-        bf16x8 xa[2], yb[2];
-#pragma unroll
-        for (int i = 0; i < 2; ++i) {
-          const int arow = wr * 32 + i * 16 + (lane & 15);
-          xa[i] = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(Xl) + wg_off(arow, k8));
-          const int brow = wc * 32 + i * 16 + (lane & 15);
-          yb[i] = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(Yl) + wg_off(brow, k8));
-        }
-#pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
-#pragma unroll
-          for (int ni = 0; ni < 2; ++ni)
-            acc[tx][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                xa[mi], yb[ni], acc[tx][mi][ni], 0, 0, 0);
-      }
-      if (tx + 1 < KW_T) __syncthreads();  // X buffer reuse for next tap
-    }
     // advance the X half's incremental decomposition by 128 px
     if (stage_x) {
       ox += 128;
@@ -207,24 +175,43 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
         if (++oy >= g.Ho) { oy = 0; ++bb; }
       }
     }
-  }
+    __syncthreads();
 
 #pragma unroll
-  for (int tx = 0; tx < KW_T; ++tx) {
+    for (int ks = 0; ks < 4; ++ks) {
+      const int k8 = (lane >> 4) + ks * 4;
+      bf16x8 xa[2], yb[2];
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+      for (int i = 0; i < 2; ++i) {
+        const int arow = wr * 32 + i * 16 + (lane & 15);
+        xa[i] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<char*>(Xl) + wg_off(arow, k8));
+        const int brow = wc * 32 + i * 16 + (lane & 15);
+        yb[i] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<char*>(Yl) + wg_off(brow, k8));
+      }
 #pragma unroll
-      for (int r2 = 0; r2 < 4; ++r2) {
-        const int ci = ci0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r2;
-        if (ci >= g.Cin) continue;
+      for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
-          if (co >= g.Cout) continue;
-          atomicAdd(
-              &dw[(((int64_t)co * g.Cin + ci) * g.KH + trow) * g.KW + tx],
-              acc[tx][mi][ni][r2]);
-        }
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              xa[mi], yb[ni], acc[mi][ni], 0, 0, 0);
+    }
+  }
+
+  const int ty = t / g.KW, tx = t % g.KW;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int r2 = 0; r2 < 4; ++r2) {
+      const int ci = ci0 + wr * 32 + mi * 16 + (lane >> 4) * 4 + r2;
+      if (ci >= g.Cin) continue;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
+        if (co >= g.Cout) continue;
+        atomicAdd(&dw[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx],
+                  acc[mi][ni][r2]);
       }
     }
   }
@@ -249,35 +236,26 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
                          xc.options().dtype(at::kFloat));
   const int ci_tiles = (int)cdiv(g.Cin, 64);
   const int co_tiles = (int)cdiv(g.Cout, 64);
-  // one block per kernel ROW of taps (KW handled in-kernel)
-  int nchunks = std::max(1,
-      (int)(1024 / std::max(1, ci_tiles * co_tiles * (int)KH)));
+  const int taps = (int)(KH * KW);
+  int nchunks = std::max(1, 1024 / (ci_tiles * co_tiles * taps));
   int chunk_len = (int)cdiv(g.M, nchunks);
   chunk_len = (int)cdiv(chunk_len, 128) * 128;
   nchunks = (int)cdiv(g.M, chunk_len);
   g.chunk_len = chunk_len;
 
-  dim3 grid(ci_tiles, co_tiles, (int)KH * nchunks);
+  dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();
   const bool aligned = (g.Cin % 8 == 0) && (g.Cout % 8 == 0);
-  auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
-  auto* pdy = reinterpret_cast<const bf16*>(dyc.data_ptr());
-  auto* pdw = dw.data_ptr<float>();
-#define RTHD_WG_LAUNCH(AL, KWV)                                              \
-  hipLaunchKernelGGL((wgrad_bf16_kernel<AL, KWV>), grid, dim3(256), 0, s,   \
-      px, pdy, pdw, g)
-  if (aligned) {
-    if (KW == 1) RTHD_WG_LAUNCH(true, 1);
-    else if (KW == 2) RTHD_WG_LAUNCH(true, 2);
-    else if (KW == 3) RTHD_WG_LAUNCH(true, 3);
-    else TORCH_CHECK(false, "wgrad_bf16_fast: unsupported KW ", KW);
-  } else {
-    if (KW == 1) RTHD_WG_LAUNCH(false, 1);
-    else if (KW == 2) RTHD_WG_LAUNCH(false, 2);
-    else if (KW == 3) RTHD_WG_LAUNCH(false, 3);
-    else TORCH_CHECK(false, "wgrad_bf16_fast: unsupported KW ", KW);
-  }
-#undef RTHD_WG_LAUNCH
+  if (aligned)
+    hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(256), 0, s,
+        reinterpret_cast<const bf16*>(xc.data_ptr()),
+        reinterpret_cast<const bf16*>(dyc.data_ptr()),
+        dw.data_ptr<float>(), g);
+  else
+    hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(256), 0, s,
+        reinterpret_cast<const bf16*>(xc.data_ptr()),
+        reinterpret_cast<const bf16*>(dyc.data_ptr()),
+        dw.data_ptr<float>(), g);
   HIP_CHECK_LAST();
   return dw;
 }

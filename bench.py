@@ -126,6 +126,7 @@ def main():
     calc = LossCalculator().to(device)
     is_bucketed = world > 1
 
+    graph_active = False
     if args.mode == 'train':
         want_graph = use_cuda and world == 1 and not args.no_train_graph
         try:
@@ -167,6 +168,7 @@ def main():
 
                 def step(i):
                     graphs[i % len(graphs)].replay()
+                graph_active = True
             except Exception as e:
                 print('train-graph capture failed (%s); eager stepping' % e)
                 step = eager_step
@@ -248,7 +250,8 @@ def main():
                 'parallelism': 'dp%d' % n_gpus,
                 'engine': args.engine,
                 'fp8': args.fp8,
-                'hipgraph': args.graph,
+                'hipgraph': (graph_active if args.mode == 'train'
+                             else args.graph),
             },
         }
         print(json.dumps(result))

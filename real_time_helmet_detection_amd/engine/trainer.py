@@ -66,6 +66,14 @@ def distributed_worker(device, ngpus_per_node, args, env_launch=False):
         init_process_group_from_args(args, rank, world_size)
 
     network, optimizer, scheduler, loss_calculator = load_network(args, dev)
+    if device == 0:
+        base = network.module if hasattr(network, 'module') else network
+        n_params = sum(p.numel() for p in base.parameters())
+        print('%s: StackedHourglass(num_stack=%d, in_ch=%d, increase_ch=%d)'
+              ': %.2fM parameters, engine=%s'
+              % (time.ctime(), args.num_stack, args.hourglass_inch,
+                 args.increase_ch, n_params / 1e6,
+                 'hip' if dev.type == 'cuda' else 'cpu-eager'))
 
     dataset = load_dataset(args)
     sampler = torch.utils.data.distributed.DistributedSampler(dataset) \

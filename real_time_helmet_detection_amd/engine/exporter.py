@@ -94,7 +94,15 @@ def build_export_module(args, network):
 
 
 def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
-    """Trace and save cpu (and, if available, gpu) TorchScript models."""
+    """Trace and save cpu (and, if available, gpu) TorchScript models.
+
+    Tracing runs with the EAGER op implementations (RTHD_EAGER_GPU=1 for the
+    GPU trace): torch.jit.trace can only record dispatcher ops, so the
+    traced graphs are pure torch-ROCm — self-contained and loadable by any
+    LibTorch (the reference's portability property, export.py:120-130). The
+    in-process inference path (Prediction/GraphedPredictor) keeps the
+    hand-written gfx950 kernels.
+    """
     import os
     predictor.eval()
     paths = {}
@@ -110,11 +118,19 @@ def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
     if do_gpu is None:
         do_gpu = torch.cuda.is_available()
     if do_gpu:
-        xg = torch.randn(1, 3, imsize, imsize, device='cuda')
-        with torch.no_grad():
-            traced_gpu = torch.jit.trace(predictor.cuda(), xg)
-        p = os.path.join(save_dir, 'jit_traced_model_gpu.pth')
-        torch.jit.save(traced_gpu, p)
-        paths['gpu'] = p
-        print('Model saved at gpu:', p)
+        prev = os.environ.get('RTHD_EAGER_GPU')
+        os.environ['RTHD_EAGER_GPU'] = '1'
+        try:
+            xg = torch.randn(1, 3, imsize, imsize, device='cuda')
+            with torch.no_grad():
+                traced_gpu = torch.jit.trace(predictor.cuda(), xg)
+            p = os.path.join(save_dir, 'jit_traced_model_gpu.pth')
+            torch.jit.save(traced_gpu, p)
+            paths['gpu'] = p
+            print('Model saved at gpu:', p)
+        finally:
+            if prev is None:
+                os.environ.pop('RTHD_EAGER_GPU', None)
+            else:
+                os.environ['RTHD_EAGER_GPU'] = prev
     return paths

@@ -94,7 +94,9 @@ int main(int argc, char** argv) {
   auto clss = out->elements()[1].toTensor().cpu();
   auto scores = out->elements()[2].toTensor().cpu();
   const char* names[] = {"hat", "person"};
-  for (int i = 0; i < boxes.size(0); ++i) {
+  const int nshow = std::min<int64_t>(boxes.size(0), 20);
+  printf("%lld detections (showing %d)\n", (long long)boxes.size(0), nshow);
+  for (int i = 0; i < nshow; ++i) {
     const int cls = clss[i].item<int64_t>();
     printf("det %2d: %-7s score %.3f box [%7.1f %7.1f %7.1f %7.1f]\n", i,
            cls < 2 ? names[cls] : "?", scores[i].item<float>(),

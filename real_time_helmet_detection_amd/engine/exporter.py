@@ -93,17 +93,21 @@ def build_export_module(args, network):
         pool_size=args.pool_size)
 
 
-def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
+def export_model(predictor, save_dir='.', imsize=512, do_gpu=None,
+                 native=True):
     """Trace and save cpu (and, if available, gpu) TorchScript models.
 
-    Tracing runs with the EAGER op implementations (RTHD_EAGER_GPU=1 for the
-    GPU trace): torch.jit.trace can only record dispatcher ops, so the
-    traced graphs are pure torch-ROCm — self-contained and loadable by any
-    LibTorch (the reference's portability property, export.py:120-130). The
-    in-process inference path (Prediction/GraphedPredictor) keeps the
-    hand-written gfx950 kernels.
+    The CPU trace records plain torch-ROCm eager ops — self-contained and
+    loadable by any LibTorch (the reference's portability property,
+    export.py:120-130). The GPU trace records the NATIVE gfx950 kernels via
+    their dispatcher registrations (torch.ops.rthd.*, TORCH_LIBRARY in
+    ops/csrc/bindings.cpp) — packed weights and folded BN scales are baked
+    into the graph as constants; the consumer (tools/cpp_infer, ``-k``)
+    dlopens the kernel extension before loading. Pass ``native=False`` (or
+    set RTHD_EXPORT_EAGER=1) for a portable eager-op GPU trace instead.
     """
     import os
+    os.makedirs(save_dir, exist_ok=True)
     predictor.eval()
     paths = {}
 
@@ -118,8 +122,11 @@ def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
     if do_gpu is None:
         do_gpu = torch.cuda.is_available()
     if do_gpu:
+        if os.environ.get('RTHD_EXPORT_EAGER'):
+            native = False
         prev = os.environ.get('RTHD_EAGER_GPU')
-        os.environ['RTHD_EAGER_GPU'] = '1'
+        if not native:
+            os.environ['RTHD_EAGER_GPU'] = '1'
         try:
             xg = torch.randn(1, 3, imsize, imsize, device='cuda')
             with torch.no_grad():
@@ -127,7 +134,8 @@ def export_model(predictor, save_dir='.', imsize=512, do_gpu=None):
             p = os.path.join(save_dir, 'jit_traced_model_gpu.pth')
             torch.jit.save(traced_gpu, p)
             paths['gpu'] = p
-            print('Model saved at gpu:', p)
+            print('Model saved at gpu (%s ops): %s'
+                  % ('native rthd' if native else 'eager', p))
         finally:
             if prev is None:
                 os.environ.pop('RTHD_EAGER_GPU', None)

@@ -101,3 +101,54 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("col_sum", &rthd::col_sum);
   m.def("tr16_probe", &rthd::tr16_probe);
 }
+
+// ---------------------------------------------------------------------------
+// Dispatcher registration (torch.ops.rthd.*) for the inference-path ops.
+//
+// torch.jit.trace can only record dispatcher ops, so the GPU export path
+// (engine/exporter.py) calls these instead of the pybind entry points: the
+// traced jit_traced_model_gpu.pth then embeds the native gfx950 kernels and
+// the C++ app (tools/cpp_infer, -k flag) runs them via LibTorch with no
+// python. The pools get tensor-only wrappers (the pybind variants return
+// [y, argmax] for autograd; inference needs just y).
+// ---------------------------------------------------------------------------
+namespace rthd {
+
+static torch::Tensor maxpool2x2_op(torch::Tensor x) {
+  return pool2x2_fwd(std::move(x), /*is_max=*/true, /*need_arg=*/false)[0];
+}
+static torch::Tensor avgpool2x2_op(torch::Tensor x) {
+  return pool2x2_fwd(std::move(x), /*is_max=*/false, /*need_arg=*/false)[0];
+}
+static torch::Tensor maxpool_same_op(torch::Tensor x, int64_t k) {
+  return maxpool_same_fwd(std::move(x), k, /*need_arg=*/false)[0];
+}
+
+}  // namespace rthd
+
+TORCH_LIBRARY(rthd, m) {
+  m.def("conv_fwd(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
+        "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
+        "int act) -> Tensor");
+  m.def("conv_fwd_fp8(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
+        "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
+        "int act) -> Tensor");
+  m.def("stem_fwd(Tensor x, Tensor w, Tensor scale, Tensor shift, "
+        "int stride, int pad, int act) -> Tensor");
+  m.def("add_act_fwd(Tensor a, Tensor b, int act) -> Tensor");
+  m.def("maxpool2x2(Tensor x) -> Tensor");
+  m.def("avgpool2x2(Tensor x) -> Tensor");
+  m.def("maxpool_same(Tensor x, int k) -> Tensor");
+  m.def("upsample2x_add(Tensor x, Tensor? skip) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
+  m.impl("conv_fwd", rthd::conv_fwd);
+  m.impl("conv_fwd_fp8", rthd::conv_fwd_fp8);
+  m.impl("stem_fwd", rthd::stem_fwd);
+  m.impl("add_act_fwd", rthd::add_act_fwd);
+  m.impl("maxpool2x2", rthd::maxpool2x2_op);
+  m.impl("avgpool2x2", rthd::avgpool2x2_op);
+  m.impl("maxpool_same", rthd::maxpool_same_op);
+  m.impl("upsample2x_add", rthd::upsample2x_add_fwd);
+}

@@ -46,6 +46,16 @@ def _C():
     return _backend.require_ext()
 
 
+def _ops():
+    """torch.ops.rthd — the dispatcher-registered views of the inference-path
+    kernels (TORCH_LIBRARY in bindings.cpp). Used for every forward call so
+    torch.jit.trace records the native ops and the exported GPU model keeps
+    the gfx950 kernels (reference export.py:120-130's portability contract,
+    upgraded to native)."""
+    _backend.require_ext()
+    return torch.ops.rthd
+
+
 def _bf16_mode(x):
     return x.dtype == torch.bfloat16 or _amp.is_autocast_enabled()
 
@@ -78,12 +88,15 @@ class _ConvBNActFn(torch.autograd.Function):
         ones = torch.ones(cout, device=dev, dtype=torch.float32)
         zeros = torch.zeros(cout, device=dev, dtype=torch.float32)
 
+        ops = _ops()
+
         def run_conv(scale, shift, act, sk=None):
             if is_stem:
                 assert sk is None
-                return C.stem_fwd(xc, weight, scale, shift, stride, pad, act)
-            return C.conv_fwd(xc, wpk, scale, shift, sk, kh, kw, stride,
-                              pad, cout, act)
+                return ops.stem_fwd(xc, weight, scale, shift, stride, pad,
+                                    act)
+            return ops.conv_fwd(xc, wpk, scale, shift, sk, kh, kw, stride,
+                                pad, cout, act)
 
         bias_f = bias.float().contiguous() if bias is not None else zeros
         mean = rstd = y_lin = None
@@ -244,8 +257,9 @@ def _conv_infer_fp8(x, conv, bn, act_code):
         scale = sw
         shift = (conv.bias.float() if conv.bias is not None
                  else torch.zeros(cout, device=x.device))
-    return C.conv_fwd_fp8(x, wpk, scale.contiguous(), shift.contiguous(),
-                          None, kh, kw, stride, pad, cout, act_code)
+    return _ops().conv_fwd_fp8(x, wpk, scale.contiguous(),
+                               shift.contiguous(), None, kh, kw, stride,
+                               pad, cout, act_code)
 
 
 def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
@@ -292,7 +306,7 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
 class _AddActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, a, b, act_code):
-        y = _C().add_act_fwd(a, b, act_code)
+        y = _ops().add_act_fwd(a, b, act_code)
         ctx.act_code = act_code
         ctx.save_for_backward(y)
         return y
@@ -320,10 +334,11 @@ class _MaxPool2x2Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
         need_arg = x.requires_grad
-        out = _C().pool2x2_fwd(x, True, need_arg)
-        if need_arg:
-            ctx.save_for_backward(out[1])
         ctx.hw = (x.shape[2], x.shape[3])
+        if not need_arg:
+            return _ops().maxpool2x2(x)
+        out = _C().pool2x2_fwd(x, True, True)
+        ctx.save_for_backward(out[1])
         return out[0]
 
     @staticmethod
@@ -336,7 +351,7 @@ class _AvgPool2x2Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
         ctx.hw = (x.shape[2], x.shape[3])
-        return _C().pool2x2_fwd(x, False, False)[0]
+        return _ops().avgpool2x2(x)
 
     @staticmethod
     def backward(ctx, dy):
@@ -348,10 +363,11 @@ class _MaxPoolSameFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, k):
         need_arg = x.requires_grad
-        out = _C().maxpool_same_fwd(x, k, need_arg)
-        if need_arg:
-            ctx.save_for_backward(out[1])
         ctx.k = k
+        if not need_arg:
+            return _ops().maxpool_same(x, k)
+        out = _C().maxpool_same_fwd(x, k, True)
+        ctx.save_for_backward(out[1])
         return out[0]
 
     @staticmethod
@@ -378,7 +394,7 @@ class _Upsample2xAddFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, skip):
         ctx.has_skip = skip is not None
-        return _C().upsample2x_add_fwd(x, skip)
+        return _ops().upsample2x_add(x, skip)
 
     @staticmethod
     def backward(ctx, dy):

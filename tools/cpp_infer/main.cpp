@@ -12,8 +12,13 @@
 // or `python tools/cpp_infer/to_ppm.py img.jpg`). Preprocessing matches
 // utils.imload: resize to 512x512 (bilinear), scale to [0,1], ImageNet
 // normalize.
+// When the traced GPU model embeds the native gfx950 ops (torch.ops.rthd.*,
+// the default export.py GPU trace), pass the kernel extension with
+// -k <path to real_time_helmet_detection_amd/ops/_C*.so>; it is dlopen'd
+// before torch::jit::load so the rthd:: custom ops resolve.
 #include <torch/script.h>
 #include <ATen/hip/HIPContext.h>
+#include <dlfcn.h>
 
 #include <chrono>
 #include <cstdio>
@@ -64,19 +69,29 @@ static torch::Tensor load_image(const std::string& path, int imsize) {
   return img;
 }
 
-int main(int argc, char** argv) {
-  std::string model_path, image_path;
+static int run(int argc, char** argv) {
+  std::string model_path, image_path, kernels_path;
   int iters = 100, imsize = 512;
   for (int i = 1; i < argc - 1; ++i) {
     if (!strcmp(argv[i], "-m")) model_path = argv[++i];
     else if (!strcmp(argv[i], "-i")) image_path = argv[++i];
+    else if (!strcmp(argv[i], "-k")) kernels_path = argv[++i];
     else if (!strcmp(argv[i], "-n")) iters = atoi(argv[++i]);
     else if (!strcmp(argv[i], "-s")) imsize = atoi(argv[++i]);
   }
   if (model_path.empty() || image_path.empty()) {
     std::cerr << "usage: " << argv[0]
-              << " -m model.pth -i image.jpg [-n iters] [-s imsize]\n";
+              << " -m model.pth -i image.ppm [-k rthd_ops.so] [-n iters]"
+                 " [-s imsize]\n";
     return 1;
+  }
+  if (!kernels_path.empty()) {
+    // the kernel extension links libtorch_python, which expects the CPython
+    // symbols to be present in the process (normally exported by the python
+    // binary) — provide them via libpython before loading the extension.
+    dlopen("libpython3.10.so.1.0", RTLD_NOW | RTLD_GLOBAL);
+    if (!dlopen(kernels_path.c_str(), RTLD_NOW | RTLD_GLOBAL))
+      throw std::runtime_error(std::string("dlopen failed: ") + dlerror());
   }
 
   torch::jit::script::Module model = torch::jit::load(model_path);
@@ -115,4 +130,13 @@ int main(int argc, char** argv) {
   printf("%d iters in %.3f s -> %.1f FPS @ %dx%d (%s)\n", iters, sec,
          iters / sec, imsize, imsize, cuda ? "gpu" : "cpu");
   return 0;
+}
+
+int main(int argc, char** argv) {
+  try {
+    return run(argc, argv);
+  } catch (const std::exception& e) {
+    std::cerr << "helmet_infer: " << e.what() << "\n";
+    return 1;
+  }
 }

@@ -40,7 +40,10 @@ struct WgradGeo2 {
   int KH, KW, stride, pad;
   int M;
   int chunk_len;
+  int ci_tiles, co_tiles, nchunks;  // XCD-clustered 1-D launch decomposition
 };
+
+#define WGRAD_NUM_XCD 8
 
 template <bool ALIGNED>  // Cin,Cout % 8 == 0: branch-free staging loads
 __global__ __launch_bounds__(256)
@@ -50,6 +53,14 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   // 4 waves; block tile [64 ci][64 co] (wave = 32 x 32). A 128x128 8-wave
   // variant measured 2.2x SLOWER per unit work (64 KB LDS -> 2 blocks/CU)
   // — this geometry keeps 5 blocks/CU resident.
+  //
+  // Chunk-major order: the (ci,co) tiles are the fastest grid dims and the
+  // 9 taps of a chunk are adjacent in z, so all taps*ci_tiles*co_tiles
+  // blocks that share one px chunk's X/dY data are CONSECUTIVE in linear
+  // dispatch order — the scheduler's locality then lets the per-XCD L2
+  // catch the tap re-reads. (An explicit XCD=L%8 clustering remap was
+  // measured 1.45x SLOWER — the dispatcher is not strict round-robin, and
+  // the remap scattered the chunk groups it was trying to cluster.)
   const int t = blockIdx.z % (g.KH * g.KW);
   const int chunk = blockIdx.z / (g.KH * g.KW);
   const int ci0 = blockIdx.x * 64;
@@ -240,8 +251,15 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   int nchunks = std::max(1, 1024 / (ci_tiles * co_tiles * taps));
   int chunk_len = (int)cdiv(g.M, nchunks);
   chunk_len = (int)cdiv(chunk_len, 128) * 128;
+  if (const char* e = getenv("RTHD_WGRAD_CHUNK")) {   // perf-tuning knob
+    const int v = atoi(e);
+    if (v >= 128) chunk_len = (int)cdiv(v, 128) * 128;
+  }
   nchunks = (int)cdiv(g.M, chunk_len);
   g.chunk_len = chunk_len;
+  g.ci_tiles = ci_tiles;
+  g.co_tiles = co_tiles;
+  g.nchunks = nchunks;
 
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();

@@ -262,6 +262,37 @@ def _conv_infer_fp8(x, conv, bn, act_code):
                                pad, cout, act_code)
 
 
+def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
+    """Plain (non-autograd.Function) inference forward: eval-BN fold into
+    the conv epilogue via torch.ops.rthd.* only — no PythonOp nodes, so
+    torch.jit.trace serializes it (export.py's native GPU trace)."""
+    ops = _ops()
+    bf16 = _bf16_mode(x) and (is_stem or x.shape[1] % 8 == 0)
+    dtype = torch.bfloat16 if bf16 else torch.float32
+    xc = x.to(dtype).contiguous(memory_format=torch.channels_last)
+    skc = None
+    if skip is not None:
+        skc = skip.to(dtype).contiguous(memory_format=torch.channels_last)
+    weight = conv.weight
+    cout = weight.shape[0]
+    bias_f = (conv.bias.float().contiguous() if conv.bias is not None
+              else torch.zeros(cout, device=x.device, dtype=torch.float32))
+    if bn is not None:
+        rstd_run = torch.rsqrt(bn.running_var.float() + bn.eps)
+        scale = (bn.weight.float() * rstd_run).contiguous()
+        shift = (bn.bias.float()
+                 + (bias_f - bn.running_mean.float()) * scale).contiguous()
+    else:
+        scale = torch.ones(cout, device=x.device, dtype=torch.float32)
+        shift = bias_f
+    if is_stem:
+        assert skc is None
+        return ops.stem_fwd(xc, weight, scale, shift, stride, pad, act_code)
+    wpk = _C().pack_weights(weight, False, bf16)
+    return ops.conv_fwd(xc, wpk, scale, shift, skc, kh, kw, stride, pad,
+                        cout, act_code)
+
+
 def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
                 skip=None):
     """GPU twin of functional.conv_bn_act (act_module path stays eager)."""
@@ -277,10 +308,13 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
     act_code = ACT_CODE.get(act)
     if act_code is None:
         raise NotImplementedError(f'HIP conv epilogue: activation {act!r}')
-    if (_amp.fp8_enabled() and not training and not is_stem
-            and not torch.is_grad_enabled() and cin % 16 == 0
-            and skip is None):
-        y = _conv_infer_fp8(x, conv, bn, act_code)
+    if not training and not torch.is_grad_enabled():
+        if (_amp.fp8_enabled() and not is_stem and cin % 16 == 0
+                and skip is None):
+            y = _conv_infer_fp8(x, conv, bn, act_code)
+        else:
+            y = _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride,
+                            pad, is_stem)
         if act_module is not None:
             y = act_module(y)
         return y
@@ -322,7 +356,10 @@ def add_act(a, b, act='Linear', act_module=None):
     code = ACT_CODE.get(act)
     if code is None:
         raise NotImplementedError(f'HIP add_act: activation {act!r}')
-    y = _AddActFn.apply(a, b, code)
+    if not torch.is_grad_enabled():
+        y = _ops().add_act_fwd(a, b, code)
+    else:
+        y = _AddActFn.apply(a, b, code)
     if act_module is not None:
         y = act_module(y)
     return y
@@ -377,14 +414,20 @@ class _MaxPoolSameFn(torch.autograd.Function):
 
 
 def maxpool2x2(x):
+    if not torch.is_grad_enabled():
+        return _ops().maxpool2x2(x)
     return _MaxPool2x2Fn.apply(x)
 
 
 def avgpool2x2(x):
+    if not torch.is_grad_enabled():
+        return _ops().avgpool2x2(x)
     return _AvgPool2x2Fn.apply(x)
 
 
 def maxpool_same(x, kernel):
+    if not torch.is_grad_enabled():
+        return _ops().maxpool_same(x, kernel)
     return _MaxPoolSameFn.apply(x, kernel)
 
 
@@ -403,6 +446,8 @@ class _Upsample2xAddFn(torch.autograd.Function):
 
 
 def upsample2x_add(x, skip=None):
+    if not torch.is_grad_enabled():
+        return _ops().upsample2x_add(x, skip)
     if skip is None:
         return _Upsample2xAddFn.apply(x, None)
     return _Upsample2xAddFn.apply(x, skip)

@@ -129,8 +129,15 @@ def export_model(predictor, save_dir='.', imsize=512, do_gpu=None,
             os.environ['RTHD_EAGER_GPU'] = '1'
         try:
             xg = torch.randn(1, 3, imsize, imsize, device='cuda')
+            predictor = predictor.cuda()
             with torch.no_grad():
-                traced_gpu = torch.jit.trace(predictor.cuda(), xg)
+                if native:
+                    # warm the conv inference caches first: the trace then
+                    # bakes the FROZEN packed-weight/scale/shift tensors as
+                    # constants (exact replay); a cold-cache trace records
+                    # the fold chain and was measured to misreplay.
+                    predictor(xg)
+                traced_gpu = torch.jit.trace(predictor, xg)
             p = os.path.join(save_dir, 'jit_traced_model_gpu.pth')
             torch.jit.save(traced_gpu, p)
             paths['gpu'] = p

@@ -95,7 +95,26 @@ __global__ void colsum8_part_kernel(const bf16* __restrict__ x,
   const int64_t r0 = blockIdx.y * rows_per_chunk;
   const int64_t r1 = min(M, r0 + rows_per_chunk);
   float acc[8] = {}, accsq[8] = {};
-  for (int64_t r = r0 + rs; r < r1; r += streams) {
+  // 4x unrolled: one 16-B load per loop iteration leaves only ~4 MB in
+  // flight across the chip (< the ~6 MB needed to cover HBM latency at
+  // 8 TB/s); four independent loads per iteration saturate the bus.
+  int64_t r = r0 + rs;
+  for (; r + 3 * (int64_t)streams < r1; r += 4 * (int64_t)streams) {
+    bf16 v[4][8];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      *reinterpret_cast<uint4*>(v[u]) = *reinterpret_cast<const uint4*>(
+          &x[(r + u * (int64_t)streams) * C + c0]);
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float f = b2f(v[u][e]);
+        acc[e] += f;
+        if (WANT_SQ) accsq[e] += f * f;
+      }
+  }
+  for (; r < r1; r += streams) {
     bf16 v[8];
     *reinterpret_cast<uint4*>(v) =
         *reinterpret_cast<const uint4*>(&x[r * C + c0]);
@@ -324,7 +343,34 @@ __global__ void bn_bwd_reduce8_part_kernel(
     bt[e] = beta[c0 + e];
   }
   float a1[8] = {}, a2[8] = {};
-  for (int64_t r = r0 + rs; r < r1; r += streams) {
+  // 2x unrolled (2-3 loads/iter already): 4-6 independent loads in flight
+  int64_t r = r0 + rs;
+  for (; r + (int64_t)streams < r1; r += 2 * (int64_t)streams) {
+    bf16 vx[2][8], vdy[2][8], vsk[2][8];
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const int64_t rr = (r + u * (int64_t)streams) * C + c0;
+      *reinterpret_cast<uint4*>(vx[u]) =
+          *reinterpret_cast<const uint4*>(&x[rr]);
+      *reinterpret_cast<uint4*>(vdy[u]) =
+          *reinterpret_cast<const uint4*>(&dy[rr]);
+      if (HAS_SKIP)
+        *reinterpret_cast<uint4*>(vsk[u]) =
+            *reinterpret_cast<const uint4*>(&skip[rr]);
+    }
+#pragma unroll
+    for (int u = 0; u < 2; ++u)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float xh = (b2f(vx[u][e]) - mu[e]) * rsd[e];
+        float pre = xh * gm[e] + bt[e];
+        if (HAS_SKIP) pre += b2f(vsk[u][e]);
+        const float dpre = b2f(vdy[u][e]) * act_grad_from_pre(pre, act);
+        a1[e] += dpre;
+        a2[e] += dpre * xh;
+      }
+  }
+  for (; r < r1; r += streams) {
     bf16 vx[8], vdy[8], vsk[8];
     *reinterpret_cast<uint4*>(vx) =
         *reinterpret_cast<const uint4*>(&x[r * C + c0]);

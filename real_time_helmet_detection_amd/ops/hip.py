@@ -265,7 +265,12 @@ def _conv_infer_fp8(x, conv, bn, act_code):
 def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
     """Plain (non-autograd.Function) inference forward: eval-BN fold into
     the conv epilogue via torch.ops.rthd.* only — no PythonOp nodes, so
-    torch.jit.trace serializes it (export.py's native GPU trace)."""
+    torch.jit.trace serializes it (export.py's native GPU trace).
+
+    Packed weights + folded scale/shift are cached on the conv module keyed
+    by the parameter/buffer versions: the serving hot path skips the
+    per-call pack + fold, and a trace bakes the frozen tensors as
+    constants."""
     ops = _ops()
     bf16 = _bf16_mode(x) and (is_stem or x.shape[1] % 8 == 0)
     dtype = torch.bfloat16 if bf16 else torch.float32
@@ -275,20 +280,35 @@ def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
         skc = skip.to(dtype).contiguous(memory_format=torch.channels_last)
     weight = conv.weight
     cout = weight.shape[0]
-    bias_f = (conv.bias.float().contiguous() if conv.bias is not None
-              else torch.zeros(cout, device=x.device, dtype=torch.float32))
-    if bn is not None:
-        rstd_run = torch.rsqrt(bn.running_var.float() + bn.eps)
-        scale = (bn.weight.float() * rstd_run).contiguous()
-        shift = (bn.bias.float()
-                 + (bias_f - bn.running_mean.float()) * scale).contiguous()
-    else:
-        scale = torch.ones(cout, device=x.device, dtype=torch.float32)
-        shift = bias_f
+
+    key = (bf16, weight._version,
+           conv.bias._version if conv.bias is not None else -1,
+           bn.weight._version if bn is not None else -1,
+           bn.bias._version if bn is not None else -1,
+           bn.running_mean._version if bn is not None else -1,
+           bn.running_var._version if bn is not None else -1)
+    cache = getattr(conv, '_rthd_infer_cache', None)
+    if cache is None or cache[0] != key:
+        bias_f = (conv.bias.float().contiguous() if conv.bias is not None
+                  else torch.zeros(cout, device=x.device,
+                                   dtype=torch.float32))
+        if bn is not None:
+            rstd_run = torch.rsqrt(bn.running_var.float() + bn.eps)
+            scale = (bn.weight.float() * rstd_run).contiguous()
+            shift = (bn.bias.float()
+                     + (bias_f - bn.running_mean.float())
+                     * scale).contiguous()
+        else:
+            scale = torch.ones(cout, device=x.device, dtype=torch.float32)
+            shift = bias_f
+        wpk = None if is_stem else _C().pack_weights(weight, False, bf16)
+        conv._rthd_infer_cache = (key, wpk, scale, shift)
+        cache = conv._rthd_infer_cache
+    _, wpk, scale, shift = cache
+
     if is_stem:
         assert skc is None
         return ops.stem_fwd(xc, weight, scale, shift, stride, pad, act_code)
-    wpk = _C().pack_weights(weight, False, bf16)
     return ops.conv_fwd(xc, wpk, scale, shift, skc, kh, kw, stride, pad,
                         cout, act_code)
 

@@ -1,5 +1,7 @@
 """Bisect the native-trace mismatch: determinism of the no-grad native path,
 no-grad path vs autograd-Function path, live vs traced."""
+import sys, os
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 from real_time_helmet_detection_amd.engine.exporter import Export, export_model
 from real_time_helmet_detection_amd.models import StackedHourglass

@@ -10,6 +10,7 @@ python. ``export_model`` traces and saves the cpu/gpu variants
 
 import torch
 
+from ..ops import _backend
 from ..transform import hm2box
 
 
@@ -81,7 +82,13 @@ class Export(torch.nn.Module):
         boxes = torch.cat(stack_boxes, dim=0)
         clss = torch.cat(stack_clss, dim=0)
         scores = torch.cat(stack_scores, dim=0)
-        keep = nms_scripted(boxes, scores, self.nms_th)
+        if boxes.is_cuda and not _backend.eager_gpu_override():
+            # single LDS-resident NMS kernel: the scripted greedy loop costs
+            # one device sync per candidate (float(s[i])) — ~50 syncs/frame
+            # in the exported C++ app
+            keep = torch.ops.rthd.nms(boxes, scores, float(self.nms_th))
+        else:
+            keep = nms_scripted(boxes, scores, self.nms_th)
         return boxes[keep], clss[keep], scores[keep]
 
 

@@ -140,6 +140,7 @@ TORCH_LIBRARY(rthd, m) {
   m.def("avgpool2x2(Tensor x) -> Tensor");
   m.def("maxpool_same(Tensor x, int k) -> Tensor");
   m.def("upsample2x_add(Tensor x, Tensor? skip) -> Tensor");
+  m.def("nms(Tensor boxes, Tensor scores, float iou) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
@@ -151,4 +152,5 @@ TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("avgpool2x2", rthd::avgpool2x2_op);
   m.impl("maxpool_same", rthd::maxpool_same_op);
   m.impl("upsample2x_add", rthd::upsample2x_add_fwd);
+  m.impl("nms", rthd::nms_fwd);
 }

@@ -61,6 +61,8 @@ class Export(torch.nn.Module):
     def forward(self, x):
         batch_output = self.network(x)  # (1, S, num_cls+4, h, w)
         outputs = batch_output[0]  # (S, num_cls+4, h, w)
+        if outputs.is_cuda and not _backend.eager_gpu_override():
+            return self._forward_native(outputs)
         stack_boxes = []
         stack_clss = []
         stack_scores = []
@@ -82,13 +84,30 @@ class Export(torch.nn.Module):
         boxes = torch.cat(stack_boxes, dim=0)
         clss = torch.cat(stack_clss, dim=0)
         scores = torch.cat(stack_scores, dim=0)
-        if boxes.is_cuda and not _backend.eager_gpu_override():
-            # single LDS-resident NMS kernel: the scripted greedy loop costs
-            # one device sync per candidate (float(s[i])) — ~50 syncs/frame
-            # in the exported C++ app
-            keep = torch.ops.rthd.nms(boxes, scores, float(self.nms_th))
-        else:
-            keep = nms_scripted(boxes, scores, self.nms_th)
+        keep = nms_scripted(boxes, scores, self.nms_th)
+        return boxes[keep], clss[keep], scores[keep]
+
+    def _forward_native(self, outputs):
+        """GPU decode: stacks-as-batch through the fused peak+topk decode
+        kernel and the LDS-resident NMS — 3 kernels instead of ~25 small
+        ops per stack plus a scripted NMS loop whose float(s[i]) costs one
+        device sync per candidate (the eager export ran at 65 FPS in the
+        C++ app; this path at 150+)."""
+        outs = outputs.float()  # (S, num_cls+4, h, w)
+        heatmap, offset, wh = outs.split([self.num_cls, 2, 2], dim=1)
+        heatmap = torch.sigmoid(heatmap)
+        if self.normalized_coord:
+            offset = torch.sigmoid(offset)
+            wh = torch.sigmoid(wh)
+        b, c, s = torch.ops.rthd.decode(
+            heatmap, offset, wh, self.scale_factor, self.topk,
+            self.pool_size, self.normalized_coord)
+        boxes = b.reshape(-1, 4)
+        clss = c.reshape(-1)
+        scores = s.reshape(-1)
+        keep = scores >= self.conf_th
+        boxes, clss, scores = boxes[keep], clss[keep], scores[keep]
+        keep = torch.ops.rthd.nms(boxes, scores, float(self.nms_th))
         return boxes[keep], clss[keep], scores[keep]
 
 

@@ -123,6 +123,13 @@ static torch::Tensor avgpool2x2_op(torch::Tensor x) {
 static torch::Tensor maxpool_same_op(torch::Tensor x, int64_t k) {
   return maxpool_same_fwd(std::move(x), k, /*need_arg=*/false)[0];
 }
+static std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> decode_op(
+    torch::Tensor hm, torch::Tensor off, torch::Tensor wh,
+    int64_t scale_factor, int64_t topk, int64_t pool_size, bool normalized) {
+  auto v = decode_fwd(std::move(hm), std::move(off), std::move(wh),
+                      scale_factor, topk, pool_size, normalized);
+  return {v[0], v[1], v[2]};
+}
 
 }  // namespace rthd
 
@@ -141,6 +148,9 @@ TORCH_LIBRARY(rthd, m) {
   m.def("maxpool_same(Tensor x, int k) -> Tensor");
   m.def("upsample2x_add(Tensor x, Tensor? skip) -> Tensor");
   m.def("nms(Tensor boxes, Tensor scores, float iou) -> Tensor");
+  m.def("decode(Tensor hm, Tensor off, Tensor wh, int scale_factor, "
+        "int topk, int pool_size, bool normalized) "
+        "-> (Tensor, Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
@@ -153,4 +163,5 @@ TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("maxpool_same", rthd::maxpool_same_op);
   m.impl("upsample2x_add", rthd::upsample2x_add_fwd);
   m.impl("nms", rthd::nms_fwd);
+  m.impl("decode", rthd::decode_op);
 }

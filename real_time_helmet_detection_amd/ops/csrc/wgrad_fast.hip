@@ -96,89 +96,95 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   int oy = rr / g.Wo;
   int ox = rr - oy * g.Wo;
 
+  // Software-pipelined staging: the global loads for chunk i+1 are ISSUED
+  // before chunk i's MFMA section, so their latency hides under the 16
+  // MFMA + LDS fragment reads; the s_waitcnt lands at the next transpose
+  // (first register use). LDS stays single-buffered — only the
+  // global->register leg is pipelined (r[8][8] carries the next chunk).
+  ushort r[8][8];
+
+#define RTHD_WG_LOADX(p0_)                                                   \
+  {                                                                          \
+    int bj = bb, oyj = oy, oxj = ox;                                         \
+    const int cbase = ci0 + s_ci;                                            \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                          \
+      const int m = (p0_) + s_px + j;                                        \
+      uint4 v;                                                               \
+      const int iy = oyj * g.stride + dyt;                                   \
+      const int ix = oxj * g.stride + dxt;                                   \
+      const bool val = m < px_end && iy >= 0 && iy < g.H && ix >= 0 &&       \
+                       ix < g.W && cbase < g.Cin;                            \
+      if (ALIGNED) {                                                         \
+        /* unconditional load from a clamped address + select-zero: a        \
+           branch around the load makes hipcc drain vmcnt(0) per element     \
+           (guide trap (c) - measured 2-6x on this kernel) */                \
+        const int64_t off = val                                              \
+            ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase          \
+            : 0;                                                             \
+        v = *reinterpret_cast<const uint4*>(x + off);                        \
+        if (!val) v = uint4{0, 0, 0, 0};                                     \
+      } else {                                                               \
+        v = uint4{0, 0, 0, 0};                                               \
+        if (val) {                                                           \
+          const ushort* src = reinterpret_cast<const ushort*>(               \
+              x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);    \
+          ushort tmp[8] = {};                                                \
+          for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];           \
+          v = *reinterpret_cast<const uint4*>(tmp);                          \
+        }                                                                    \
+      }                                                                      \
+      *reinterpret_cast<uint4*>(r[j]) = v;                                   \
+      if (++oxj >= g.Wo) {                                                   \
+        oxj = 0;                                                             \
+        if (++oyj >= g.Ho) { oyj = 0; ++bj; }                                \
+      }                                                                      \
+    }                                                                        \
+  }
+
+#define RTHD_WG_LOADY(p0_)                                                   \
+  {                                                                          \
+    const int cbase = co0 + s_ci;                                            \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                          \
+      const int m = (p0_) + s_px + j;                                        \
+      uint4 v;                                                               \
+      const bool val = m < px_end && cbase < g.Cout;                         \
+      if (ALIGNED) {                                                         \
+        const int64_t off = val ? (int64_t)m * g.Cout + cbase : 0;           \
+        v = *reinterpret_cast<const uint4*>(dy + off);                       \
+        if (!val) v = uint4{0, 0, 0, 0};                                     \
+      } else {                                                               \
+        v = uint4{0, 0, 0, 0};                                               \
+        if (val) {                                                           \
+          const ushort* src = reinterpret_cast<const ushort*>(               \
+              dy + (int64_t)m * g.Cout + cbase);                             \
+          ushort tmp[8] = {};                                                \
+          for (int e = 0; cbase + e < g.Cout; ++e) tmp[e] = src[e];          \
+          v = *reinterpret_cast<const uint4*>(tmp);                          \
+        }                                                                    \
+      }                                                                      \
+      *reinterpret_cast<uint4*>(r[j]) = v;                                   \
+    }                                                                        \
+  }
+
+  // prologue: load the first chunk
+  if (stage_x) RTHD_WG_LOADX(px_start) else RTHD_WG_LOADY(px_start)
+
   for (int p0 = px_start; p0 < px_end; p0 += 128) {
-    __syncthreads();
-    ushort r[8][8];
-    if (stage_x) {
-      int bj = bb, oyj = oy, oxj = ox;
-      const int cbase = ci0 + s_ci;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int m = p0 + s_px + j;
-        uint4 v;
-        const int iy = oyj * g.stride + dyt;
-        const int ix = oxj * g.stride + dxt;
-        const bool val = m < px_end && iy >= 0 && iy < g.H && ix >= 0 &&
-                         ix < g.W && cbase < g.Cin;
-        if (ALIGNED) {
-          // UNCONDITIONAL load from a clamped address + select-zero: a
-          // branch around the load makes hipcc drain vmcnt(0) per element
-          // (guide §5 trap (c) — measured 2-6x on this kernel)
-          const int64_t off = val
-              ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase
-              : 0;
-          v = *reinterpret_cast<const uint4*>(x + off);
-          if (!val) v = uint4{0, 0, 0, 0};
-        } else {
-          v = uint4{0, 0, 0, 0};
-          if (val) {
-            const ushort* src = reinterpret_cast<const ushort*>(
-                x + (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase);
-            ushort tmp[8] = {};
-            for (int e = 0; cbase + e < g.Cin; ++e) tmp[e] = src[e];
-            v = *reinterpret_cast<const uint4*>(tmp);
-          }
-        }
-        *reinterpret_cast<uint4*>(r[j]) = v;
-        if (++oxj >= g.Wo) {
-          oxj = 0;
-          if (++oyj >= g.Ho) { oyj = 0; ++bj; }
-        }
-      }
+    __syncthreads();   // close the previous chunk's fragment reads
+    // transpose r (8 px x 8 ch) -> LDS [ch][px]
+    {
+      bf16* half = stage_x ? Xl : Yl;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         ushort o[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) o[j] = r[j][e];
         *reinterpret_cast<uint4*>(
-            reinterpret_cast<char*>(Xl) +
-            wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
-      }
-    } else {
-      const int cbase = co0 + s_ci;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int m = p0 + s_px + j;
-        uint4 v;
-        const bool val = m < px_end && cbase < g.Cout;
-        if (ALIGNED) {
-          const int64_t off = val ? (int64_t)m * g.Cout + cbase : 0;
-          v = *reinterpret_cast<const uint4*>(dy + off);
-          if (!val) v = uint4{0, 0, 0, 0};
-        } else {
-          v = uint4{0, 0, 0, 0};
-          if (val) {
-            const ushort* src =
-                reinterpret_cast<const ushort*>(dy + (int64_t)m * g.Cout +
-                                                cbase);
-            ushort tmp[8] = {};
-            for (int e = 0; cbase + e < g.Cout; ++e) tmp[e] = src[e];
-            v = *reinterpret_cast<const uint4*>(tmp);
-          }
-        }
-        *reinterpret_cast<uint4*>(r[j]) = v;
-      }
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        ushort o[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) o[j] = r[j][e];
-        *reinterpret_cast<uint4*>(
-            reinterpret_cast<char*>(Yl) +
+            reinterpret_cast<char*>(half) +
             wg_off(s_ci + e, s_px >> 3)) = *reinterpret_cast<uint4*>(o);
       }
     }
-    // advance the X half's incremental decomposition by 128 px
+    // advance the X half's incremental decomposition to p0+128
     if (stage_x) {
       ox += 128;
       while (ox >= g.Wo) {
@@ -186,7 +192,11 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
         if (++oy >= g.Ho) { oy = 0; ++bb; }
       }
     }
-    __syncthreads();
+    __syncthreads();   // publish the tile
+    // issue NEXT chunk's loads now; the wait lands at the next transpose
+    if (p0 + 128 < px_end) {
+      if (stage_x) RTHD_WG_LOADX(p0 + 128) else RTHD_WG_LOADY(p0 + 128)
+    }
 
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
@@ -209,6 +219,9 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
               xa[mi], yb[ni], acc[mi][ni], 0, 0, 0);
     }
   }
+
+#undef RTHD_WG_LOADX
+#undef RTHD_WG_LOADY
 
   const int ty = t / g.KW, tx = t % g.KW;
 #pragma unroll

@@ -60,6 +60,22 @@ def _bf16_mode(x):
     return x.dtype == torch.bfloat16 or _amp.is_autocast_enabled()
 
 
+# per-(device, n) constant epilogue vectors: the training path needs
+# ones/zeros per conv call — allocating them fresh was ~130 tiny kernel
+# launches per train step
+_const_vecs = {}
+
+
+def _ones_zeros(n, device):
+    key = (device.index if device.index is not None else -1, n)
+    v = _const_vecs.get(key)
+    if v is None:
+        v = (torch.ones(n, device=device, dtype=torch.float32),
+             torch.zeros(n, device=device, dtype=torch.float32))
+        _const_vecs[key] = v
+    return v
+
+
 # ------------------------------------------------------------------ conv ---
 
 class _ConvBNActFn(torch.autograd.Function):
@@ -85,8 +101,7 @@ class _ConvBNActFn(torch.autograd.Function):
         if not is_stem:
             wpk = C.pack_weights(weight, False, bf16)
 
-        ones = torch.ones(cout, device=dev, dtype=torch.float32)
-        zeros = torch.zeros(cout, device=dev, dtype=torch.float32)
+        ones, zeros = _ones_zeros(cout, dev)
 
         ops = _ops()
 
@@ -197,8 +212,7 @@ class _ConvBNActFn(torch.autograd.Function):
                     'stem dgrad (3-channel input) is not needed: the image '
                     'is a leaf tensor')
             cin = xc.shape[1]
-            ones = torch.ones(cin, device=xc.device, dtype=torch.float32)
-            zeros = torch.zeros(cin, device=xc.device, dtype=torch.float32)
+            ones, zeros = _ones_zeros(cin, xc.device)
             if stride == 1:
                 bf16_d = bf16 and dpre.shape[1] % 8 == 0
                 wpk_t = C.pack_weights(weight, True, bf16_d)

@@ -52,6 +52,8 @@ torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
                        int64_t stride, int64_t pad, int64_t act);
 torch::Tensor stem_wgrad(torch::Tensor x, torch::Tensor dy, int64_t stride,
                          int64_t pad);
+torch::Tensor stem_im2col(torch::Tensor x, int64_t KS, int64_t stride,
+                          int64_t pad);
 torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
                     int64_t KW, int64_t stride, int64_t pad);
 torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
@@ -93,6 +95,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("stem_fwd", &rthd::stem_fwd);
   m.def("stem_wgrad", &rthd::stem_wgrad);
+  m.def("stem_im2col", &rthd::stem_im2col);
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
   m.def("bn_stats", &rthd::bn_stats);

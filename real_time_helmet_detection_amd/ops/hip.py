@@ -185,7 +185,7 @@ class _ConvBNActFn(torch.autograd.Function):
             with torch.cuda.stream(side):
                 if is_stem and xc.dtype == torch.bfloat16 and kh == 7:
                     # im2col the 3-channel input to [px][152] rows and run
-                    # the MFMA wgrad as its 1x1 case (~4.5x faster than the
+                    # the MFMA wgrad as its 1x1 case (~2.4x faster than the
                     # direct VALU stem_wgrad); column t*3+ci -> dW[co][ci][t]
                     xcol = C.stem_im2col(xc, kh, stride, pad)
                     cout = dpre.shape[1]

@@ -81,7 +81,13 @@ def bench_stem(iters):
     dy = torch.randn(16, 64, 256, 256, device='cuda',
                      dtype=torch.bfloat16).contiguous(memory_format=CL)
     ms = timeit(lambda: C.stem_wgrad(x, dy, 2, 3), iters)
-    print(f'stem_wgrad @512^2 B16         {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
+    print(f'stem_wgrad(direct) @512^2 B16 {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
+
+    def im2col_path():
+        xcol = C.stem_im2col(x, 7, 2, 3)
+        return C.wgrad_bf16_fast(xcol, dy, 1, 1, 1, 0)
+    ms = timeit(im2col_path, iters)
+    print(f'stem_wgrad(im2col) @512^2 B16 {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
 
 
 def bench_bn(iters):

@@ -223,7 +223,12 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
 #undef RTHD_WG_LOADX
 #undef RTHD_WG_LOADY
 
+  // Per-chunk PARTIAL output (plain stores; each (chunk,tap,ci,co) cell is
+  // owned by exactly one block): summed by wgrad_reduce_kernel in a fixed
+  // order — the previous atomicAdd writeback made weight gradients vary
+  // run to run (fp add order), which amplifies at depth.
   const int ty = t / g.KW, tx = t % g.KW;
+  float* dwc = dw + (int64_t)chunk * g.Cout * g.Cin * g.KH * g.KW;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
@@ -234,10 +239,34 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
       for (int ni = 0; ni < 2; ++ni) {
         const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
-        atomicAdd(&dw[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx],
-                  acc[mi][ni][r2]);
+        dwc[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx] =
+            acc[mi][ni][r2];
       }
     }
+  }
+}
+
+// out[n] = sum_k part[k][n] — deterministic fixed-order reduction.
+__global__ void wgrad_reduce_kernel(const float* __restrict__ part,
+                                    float* __restrict__ out,
+                                    int K, int64_t N) {
+  const int64_t n4 = N >> 2;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < n4; i += (int64_t)gridDim.x * blockDim.x) {
+    float4 a = {0.f, 0.f, 0.f, 0.f};
+    for (int k = 0; k < K; ++k) {
+      const float4 v =
+          *reinterpret_cast<const float4*>(&part[(int64_t)k * N + i * 4]);
+      a.x += v.x; a.y += v.y; a.z += v.z; a.w += v.w;
+    }
+    *reinterpret_cast<float4*>(&out[i * 4]) = a;
+  }
+  // scalar tail (N % 4)
+  for (int64_t i = n4 * 4 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < N; i += (int64_t)gridDim.x * blockDim.x) {
+    float a = 0.f;
+    for (int k = 0; k < K; ++k) a += part[(int64_t)k * N + i];
+    out[i] = a;
   }
 }
 
@@ -256,8 +285,6 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
   g.M = g.B * g.Ho * g.Wo;
 
-  auto dw = torch::zeros({g.Cout, g.Cin, KH, KW},
-                         xc.options().dtype(at::kFloat));
   const int ci_tiles = (int)cdiv(g.Cin, 64);
   const int co_tiles = (int)cdiv(g.Cout, 64);
   const int taps = (int)(KH * KW);
@@ -278,6 +305,12 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   g.co_tiles = co_tiles;
   g.nchunks = nchunks;
 
+  const int64_t N = (int64_t)g.Cout * g.Cin * KH * KW;
+  auto dwp = torch::empty({(int64_t)nchunks * N},
+                          xc.options().dtype(at::kFloat));
+  auto dw = torch::empty({g.Cout, g.Cin, KH, KW},
+                         xc.options().dtype(at::kFloat));
+
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();
   const bool aligned = (g.Cin % 8 == 0) && (g.Cout % 8 == 0);
@@ -285,12 +318,15 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
     hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(256), 0, s,
         reinterpret_cast<const bf16*>(xc.data_ptr()),
         reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        dw.data_ptr<float>(), g);
+        dwp.data_ptr<float>(), g);
   else
     hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(256), 0, s,
         reinterpret_cast<const bf16*>(xc.data_ptr()),
         reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        dw.data_ptr<float>(), g);
+        dwp.data_ptr<float>(), g);
+  const int rblocks = (int)std::min<int64_t>(2048, cdiv(N, 4 * 256) + 1);
+  hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(rblocks), dim3(256), 0, s,
+      dwp.data_ptr<float>(), dw.data_ptr<float>(), nchunks, N);
   HIP_CHECK_LAST();
   return dw;
 }

@@ -402,3 +402,20 @@ def test_stem_wgrad(dtype):
     got = _C().stem_wgrad(to_gpu(x, dtype), to_gpu(dy, dtype), 2, 3)
     tol = 0.02 if dtype == torch.bfloat16 else 1e-4
     assert rel_err(got, w.grad) < tol
+
+
+@pytest.mark.gpu
+def test_wgrad_deterministic():
+    """Per-chunk-partial writeback: identical inputs must give BITWISE
+    identical weight gradients across runs (the atomicAdd version varied
+    with fp add order)."""
+    from real_time_helmet_detection_amd.ops import _backend
+    C = _backend.require_ext()
+    torch.manual_seed(0)
+    x = torch.randn(4, 64, 32, 32, device='cuda', dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    dy = torch.randn(4, 64, 32, 32, device='cuda', dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    a = C.wgrad_bf16_fast(x, dy, 3, 3, 1, 1)
+    b = C.wgrad_bf16_fast(x, dy, 3, 3, 1, 1)
+    assert torch.equal(a, b)

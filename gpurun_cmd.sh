@@ -2,9 +2,7 @@ set -x
 cd /root/repo
 export TMPDIR=/tmp
 mkdir -p gpurun_out
-timeout 600 python -m pytest tests/test_gpu_kernels.py tests/test_gpu_e2e.py -x -q -k "wgrad or conv or backward or overfit" > gpurun_out/pytest_k.log 2>&1
-echo "P_RC=$?"; tail -2 gpurun_out/pytest_k.log
-timeout 300 python tools/kbench.py wgrad --iters 30 > gpurun_out/kb_w.log 2>&1
-grep wgrad gpurun_out/kb_w.log
+timeout 900 python -m pytest tests -m gpu -x -q > gpurun_out/pytest_gpu.log 2>&1
+echo "PG_RC=$?"; tail -2 gpurun_out/pytest_gpu.log
 timeout 600 python bench.py --steps 30 --warmup 10 > gpurun_out/bench_train.json 2> gpurun_out/bench_train.log
 echo "BT_RC=$?"; cat gpurun_out/bench_train.json

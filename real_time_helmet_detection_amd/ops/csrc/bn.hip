@@ -153,11 +153,12 @@ __global__ void colsum8_part_kernel(const bf16* __restrict__ x,
   }
 }
 
-// out[c] = sum_k p[k][c]  (and out2/p2 when given).
+// stage1[ks][c] = sum of the ks-th K-slice of p[k][c] (and stage2/p2).
 // block = 64 channels x 4 k-substreams (a K=256 serial loop in one tiny
-// block was 44 us — latency-bound).
-// grid (C/64, KSPLIT): each block sums its K-slice, atomically adds into
-// the zero-initialized outputs (KSPLIT same-address atomics: negligible)
+// block was 44 us — latency-bound). grid (C/64, KSPLIT); a tiny second
+// kernel sums the KSPLIT staged rows in FIXED order (the earlier
+// same-address atomicAdd finish made BN statistics — and so the whole
+// training trajectory — vary with fp add order).
 __global__ void reduce_partials_kernel(const float* __restrict__ p1,
                                        const float* __restrict__ p2,
                                        float* __restrict__ o1,
@@ -181,13 +182,44 @@ __global__ void reduce_partials_kernel(const float* __restrict__ p1,
   if (sub == 0 && c < C) {
     a = sha[threadIdx.x] + sha[threadIdx.x + 64] + sha[threadIdx.x + 128] +
         sha[threadIdx.x + 192];
-    atomicAdd(&o1[c], a);
+    o1[(int64_t)blockIdx.y * C + c] = a;
     if (p2) {
       b = shb[threadIdx.x] + shb[threadIdx.x + 64] + shb[threadIdx.x + 128] +
           shb[threadIdx.x + 192];
-      atomicAdd(&o2[c], b);
+      o2[(int64_t)blockIdx.y * C + c] = b;
     }
   }
+}
+
+__global__ void reduce_final_kernel(const float* __restrict__ s1,
+                                    const float* __restrict__ s2,
+                                    float* __restrict__ o1,
+                                    float* __restrict__ o2, int K, int C) {
+  const int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f;
+  for (int k = 0; k < K; ++k) a += s1[(int64_t)k * C + c];
+  o1[c] = a;
+  if (o2) {
+    float b = 0.f;
+    for (int k = 0; k < K; ++k) b += s2[(int64_t)k * C + c];
+    o2[c] = b;
+  }
+}
+
+// staged two-kernel deterministic reduction of [chunks][C] partials
+static void run_reduce_partials(const float* p1, const float* p2,
+                                float* o1, float* o2, int chunks, int C,
+                                const torch::TensorOptions& opt,
+                                hipStream_t s) {
+  const int ks = std::min(8, chunks);
+  auto st = torch::empty({(int64_t)2 * ks * C}, opt);
+  float* s1 = st.data_ptr<float>();
+  float* s2 = p2 ? s1 + (int64_t)ks * C : nullptr;
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64), ks),
+      dim3(256), 0, s, p1, p2, s1, s2, chunks, C);
+  hipLaunchKernelGGL(reduce_final_kernel, dim3(cdiv(C, 256)), dim3(256),
+      0, s, s1, s2, o1, o2, ks, C);
 }
 
 // finalize mean/rstd (+ running-stat update, torch semantics)
@@ -533,10 +565,9 @@ static void run_colsum(const torch::Tensor& xc, torch::Tensor& sum,
             p1.data_ptr<float>(), p2p, M, C);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel,
-      dim3(cdiv(C, 64), std::min(8, chunks)), dim3(256),
-      0, s, p1.data_ptr<float>(), p2p, sum.data_ptr<float>(),
-      sumsq ? sumsq->data_ptr<float>() : nullptr, chunks, C);
+  run_reduce_partials(p1.data_ptr<float>(), p2p, sum.data_ptr<float>(),
+                      sumsq ? sumsq->data_ptr<float>() : nullptr, chunks,
+                      C, sum.options(), s);
 }
 
 std::vector<torch::Tensor> bn_stats(torch::Tensor x,
@@ -687,10 +718,9 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
             p1.data_ptr<float>(), p2.data_ptr<float>(), M, C, (int)act);
     });
   }
-  hipLaunchKernelGGL(reduce_partials_kernel,
-      dim3(cdiv(C, 64), std::min(8, chunks)), dim3(256),
-      0, s, p1.data_ptr<float>(), p2.data_ptr<float>(),
-      s1.data_ptr<float>(), s2.data_ptr<float>(), chunks, C);
+  run_reduce_partials(p1.data_ptr<float>(), p2.data_ptr<float>(),
+                      s1.data_ptr<float>(), s2.data_ptr<float>(), chunks,
+                      C, s1.options(), s);
 
   if (fast8_ok(xc, C)) {
     const int octs = C / 8;

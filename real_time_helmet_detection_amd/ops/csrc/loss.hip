@@ -25,7 +25,7 @@ __global__ void centernet_loss_sums_kernel(
     const float* __restrict__ poff, const float* __restrict__ goff,
     const float* __restrict__ psize, const float* __restrict__ gsize,
     const float* __restrict__ mask,
-    float* __restrict__ sums,  // [NSUMS], zero-initialized
+    float* __restrict__ part,  // [gridDim.x][NSUMS] per-block partials
     int B, int C, int64_t HW, float alpha, float beta) {
   const float eps = 1e-7f;
   float a_pos = 0.f, a_neg = 0.f, a_off = 0.f, a_size = 0.f, a_np = 0.f;
@@ -67,19 +67,31 @@ __global__ void centernet_loss_sums_kernel(
   for (int k = 0; k < NSUMS; ++k) {
     __syncthreads();
     const float r = block_reduce_sum(vals[k], smem);
-    if (threadIdx.x == 0) atomicAdd(&sums[k], r);
+    // plain per-block partial store (atomicAdd made the loss value — and
+    // through `sums`, the gradients — vary with fp add order)
+    if (threadIdx.x == 0) part[blockIdx.x * NSUMS + k] = r;
   }
 }
 
-// losses[0..2] = hm, off, size
-__global__ void centernet_loss_final_kernel(const float* __restrict__ sums,
+// reduces the [nblk][NSUMS] partials in fixed order into sums[NSUMS],
+// then losses[0..2] = hm, off, size
+__global__ void centernet_loss_final_kernel(const float* __restrict__ part,
+                                            float* __restrict__ sums,
                                             float* __restrict__ losses,
-                                            int B) {
-  const float np = fminf(fmaxf(sums[S_NPOS], 1.f), 1e30f);
-  const float inv = 1.f / ((float)B * np);
-  losses[0] = -(sums[S_POS] + sums[S_NEG]) * inv;
-  losses[1] = sums[S_OFF] * inv;
-  losses[2] = sums[S_SIZE] * inv;
+                                            int nblk, int B) {
+  if (threadIdx.x < NSUMS) {
+    float a = 0.f;
+    for (int k = 0; k < nblk; ++k) a += part[k * NSUMS + threadIdx.x];
+    sums[threadIdx.x] = a;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const float np = fminf(fmaxf(sums[S_NPOS], 1.f), 1e30f);
+    const float inv = 1.f / ((float)B * np);
+    losses[0] = -(sums[S_POS] + sums[S_NEG]) * inv;
+    losses[1] = sums[S_OFF] * inv;
+    losses[2] = sums[S_SIZE] * inv;
+  }
 }
 
 __global__ void centernet_loss_bwd_kernel(
@@ -149,22 +161,23 @@ std::vector<torch::Tensor> centernet_loss_fwd(
   const int B = phm_.size(0), C = phm_.size(1);
   const int64_t HW = (int64_t)phm_.size(2) * phm_.size(3);
 
-  auto sums = torch::zeros({NSUMS}, phm_.options());
+  auto sums = torch::empty({NSUMS}, phm_.options());
   auto losses = torch::empty({3}, phm_.options());
   auto s = at::cuda::getCurrentCUDAStream();
   const int64_t n = (int64_t)B * C * HW;
-  // cap blocks: every block ends with 5 same-address atomics (guide G12)
   int nblk = ew_grid(n, 256);
   if (nblk > 512) nblk = 512;
+  auto part = torch::empty({(int64_t)nblk * NSUMS}, phm_.options());
   hipLaunchKernelGGL(centernet_loss_sums_kernel, dim3(nblk),
       dim3(256), 0, s,
       phm_.data_ptr<float>(), ghm_.data_ptr<float>(),
       poff_.data_ptr<float>(), goff_.data_ptr<float>(),
       psize_.data_ptr<float>(), gsize_.data_ptr<float>(),
-      mask_.data_ptr<float>(), sums.data_ptr<float>(),
+      mask_.data_ptr<float>(), part.data_ptr<float>(),
       B, C, HW, (float)alpha, (float)beta);
-  hipLaunchKernelGGL(centernet_loss_final_kernel, dim3(1), dim3(1), 0, s,
-      sums.data_ptr<float>(), losses.data_ptr<float>(), B);
+  hipLaunchKernelGGL(centernet_loss_final_kernel, dim3(1), dim3(64), 0, s,
+      part.data_ptr<float>(), sums.data_ptr<float>(),
+      losses.data_ptr<float>(), nblk, B);
   HIP_CHECK_LAST();
   return {losses, sums};
 }

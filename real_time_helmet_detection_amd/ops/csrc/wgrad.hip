@@ -162,8 +162,9 @@ void wgrad_kernel(const T* __restrict__ x, const T* __restrict__ dy,
     }
   }
 
-  // ---- epilogue: atomicAdd into unpacked dW ----
+  // ---- epilogue: per-chunk partials (deterministic; see wgrad_fast) ----
   const int ty = t / g.KW, tx = t % g.KW;
+  float* dwc = dw + (int64_t)chunk * g.Cout * g.Cin * g.KH * g.KW;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
@@ -174,11 +175,23 @@ void wgrad_kernel(const T* __restrict__ x, const T* __restrict__ dy,
       for (int ni = 0; ni < 2; ++ni) {
         const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
-        const float v = acc[mi][ni][r];
-        atomicAdd(&dw[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx],
-                  v);
+        dwc[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx] =
+            acc[mi][ni][r];
       }
     }
+  }
+}
+
+// fixed-order sum over the [K][N] partials (duplicated from wgrad_fast —
+// cross-TU kernel launches need RDC)
+__global__ void wgrad_reduce_f32_kernel(const float* __restrict__ part,
+                                        float* __restrict__ out,
+                                        int K, int64_t N) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < N; i += (int64_t)gridDim.x * blockDim.x) {
+    float a = 0.f;
+    for (int k = 0; k < K; ++k) a += part[(int64_t)k * N + i];
+    out[i] = a;
   }
 }
 
@@ -197,8 +210,6 @@ torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
   g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
   g.M = g.B * g.Ho * g.Wo;
 
-  auto dw = torch::zeros({g.Cout, g.Cin, KH, KW},
-                         xc.options().dtype(at::kFloat));
   const int ci_tiles = (int)cdiv(g.Cin, 64);
   const int co_tiles = (int)cdiv(g.Cout, 64);
   const int taps = (int)(KH * KW);
@@ -211,19 +222,28 @@ torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
   g.chunk_len = chunk_len;
   g.nchunks = nchunks;
 
+  const int64_t N = (int64_t)g.Cout * g.Cin * KH * KW;
+  auto dwp = torch::empty({(int64_t)nchunks * N},
+                          xc.options().dtype(at::kFloat));
+  auto dw = torch::empty({g.Cout, g.Cin, KH, KW},
+                         xc.options().dtype(at::kFloat));
+
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();
   if (xc.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL((wgrad_kernel<bf16>), grid, dim3(256), 0, s,
         reinterpret_cast<const bf16*>(xc.data_ptr()),
         reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        dw.data_ptr<float>(), g);
+        dwp.data_ptr<float>(), g);
   } else {
     TORCH_CHECK(xc.scalar_type() == at::kFloat);
     hipLaunchKernelGGL((wgrad_kernel<float>), grid, dim3(256), 0, s,
         xc.data_ptr<float>(), dyc.data_ptr<float>(),
-        dw.data_ptr<float>(), g);
+        dwp.data_ptr<float>(), g);
   }
+  const int rblocks = (int)std::min<int64_t>(2048, cdiv(N, 256) + 1);
+  hipLaunchKernelGGL(wgrad_reduce_f32_kernel, dim3(rblocks), dim3(256),
+      0, s, dwp.data_ptr<float>(), dw.data_ptr<float>(), nchunks, N);
   HIP_CHECK_LAST();
   return dw;
 }

@@ -316,13 +316,23 @@ void stem_im2col_kernel(const bf16* __restrict__ x, bf16* __restrict__ out,
   for (int ox = threadIdx.x; ox < Wo; ox += blockDim.x) {
     bf16* row = out + ((int64_t)(b * Ho + oy) * Wo + ox) * Cp;
     const int x0 = ox * stride;                  // ix = x0 + tx - pad + pad
-    int c = 0;
-    for (int ty = 0; ty < KS; ++ty) {
-      const bf16* src = xs + (ty * span + x0) * STEM_CIN;
+    // build 8-element groups in registers and store as b128: 152 scalar
+    // 2-B global stores per px were the kernel's instruction bottleneck
+    const int ROWLEN = KS * STEM_CIN;            // 21 cols per ty
+    for (int gbase = 0; gbase < Cp; gbase += 8) {
+      ushort o[8];
 #pragma unroll
-      for (int e = 0; e < KS * STEM_CIN; ++e) row[c++] = src[e];
+      for (int k = 0; k < 8; ++k) {
+        const int c = gbase + k;
+        const int ty = c / ROWLEN;
+        const int e = c - ty * ROWLEN;
+        o[k] = (c < KS * ROWLEN)
+                   ? reinterpret_cast<const ushort*>(
+                         xs + (ty * span + x0) * STEM_CIN)[e]
+                   : (ushort)0;
+      }
+      *reinterpret_cast<uint4*>(row + gbase) = *reinterpret_cast<uint4*>(o);
     }
-    for (; c < Cp; ++c) row[c] = bf16(0.0f);
   }
 }
 

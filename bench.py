@@ -243,8 +243,9 @@ def main():
             'dtype': args.dtype if use_cuda else 'fp32',
             'data': 'synthetic',
             'config': {
-                'model': 'hourglass-%d-ch%d' % (args.num_stack,
-                                                args.hourglass_inch),
+                'model': 'hourglass-%d-ch%d%s' % (
+                    args.num_stack, args.hourglass_inch,
+                    '-inc%d' % args.increase_ch if args.increase_ch else ''),
                 'global_batch': args.batch_size * n_gpus,
                 'imsize': args.imsize,
                 'parallelism': 'dp%d' % n_gpus,

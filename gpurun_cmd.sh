@@ -1,8 +1,8 @@
-cd /root/repo
+cd /tmp
 export TMPDIR=/tmp
-mkdir -p gpurun_out
-timeout 700 python -m pytest tests -m gpu -q > gpurun_out/pytest_gpu.log 2>&1
-echo "PG_RC=$?"; tail -1 gpurun_out/pytest_gpu.log
-timeout 240 python -c "import __graft_entry__ as g; g.smoke(); print('SMOKE-OK')" 2>&1 | tail -1
-timeout 400 python bench.py --steps 30 --warmup 10 > gpurun_out/bench_train.json 2> gpurun_out/bench_train.log
-echo "BT_RC=$?"; cat gpurun_out/bench_train.json
+mkdir -p /root/repo/gpurun_out
+timeout 500 python /root/repo/bench.py --steps 10 --warmup 3 --num-stack 2 --increase-ch 128 > /root/repo/gpurun_out/bench_big.json 2> /root/repo/gpurun_out/bench_big.log
+echo "BIG_RC=$?"; cat /root/repo/gpurun_out/bench_big.json; tail -2 /root/repo/gpurun_out/bench_big.log
+timeout 500 rocprofv3 --kernel-trace --stats -d /tmp/prof3 -- python /root/repo/bench.py --steps 4 --warmup 2 --no-train-graph > /root/repo/gpurun_out/prof3.log 2>&1
+echo "PROF_RC=$?"
+cp -r /tmp/prof3 /root/repo/gpurun_out/prof3 2>/dev/null

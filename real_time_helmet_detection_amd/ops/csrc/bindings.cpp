@@ -154,6 +154,7 @@ TORCH_LIBRARY(rthd, m) {
   m.def("decode(Tensor hm, Tensor off, Tensor wh, int scale_factor, "
         "int topk, int pool_size, bool normalized) "
         "-> (Tensor, Tensor, Tensor)");
+  m.def("stem_im2col(Tensor x, int ks, int stride, int pad) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
@@ -167,4 +168,5 @@ TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("upsample2x_add", rthd::upsample2x_add_fwd);
   m.impl("nms", rthd::nms_fwd);
   m.impl("decode", rthd::decode_op);
+  m.impl("stem_im2col", rthd::stem_im2col);
 }

@@ -78,6 +78,17 @@ def _ones_zeros(n, device):
 
 # ------------------------------------------------------------------ conv ---
 
+def _stem_col_weight(weight):
+    """(Cout,3,KS,KS) -> (Cout, KS*KS*3 padded to x8, 1, 1): column t*3+ci,
+    matching stem_im2col's unfolded layout (7x7: 147 -> 152 cols)."""
+    cout = weight.shape[0]
+    w = weight.permute(0, 2, 3, 1).reshape(cout, -1)
+    padc = (-w.shape[1]) % 8
+    if padc:
+        w = torch.nn.functional.pad(w, (0, padc))
+    return w.reshape(cout, -1, 1, 1)
+
+
 class _ConvBNActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, gamma, beta, rmean, rvar, skip,
@@ -97,8 +108,17 @@ class _ConvBNActFn(torch.autograd.Function):
         cout = weight.shape[0]
         dev = x.device
 
+        # bf16 stem: unfold to [px][152] once and run the MFMA conv as its
+        # 1x1 case — both faster than the direct VALU stem kernel AND the
+        # unfolded tensor is reused by the backward wgrad (which otherwise
+        # re-unfolds). fp32 stem keeps the direct kernel.
+        stem_col = is_stem and bf16 and kh == 7
         wpk = None
-        if not is_stem:
+        if stem_col:
+            xcol = C.stem_im2col(xc, kh, stride, pad)
+            wpk = C.pack_weights(_stem_col_weight(weight), False, True)
+            xc = xcol
+        elif not is_stem:
             wpk = C.pack_weights(weight, False, bf16)
 
         ones, zeros = _ones_zeros(cout, dev)
@@ -106,6 +126,10 @@ class _ConvBNActFn(torch.autograd.Function):
         ops = _ops()
 
         def run_conv(scale, shift, act, sk=None):
+            if stem_col:
+                assert sk is None
+                return ops.conv_fwd(xc, wpk, scale, shift, None, 1, 1, 1,
+                                    0, cout, act)
             if is_stem:
                 assert sk is None
                 return ops.stem_fwd(xc, weight, scale, shift, stride, pad,
@@ -131,7 +155,8 @@ class _ConvBNActFn(torch.autograd.Function):
             y = run_conv(ones, bias_f, act_code, skc)
 
         ctx.meta = (kh, kw, stride, pad, act_code, use_bn, training, bf16,
-                    is_stem, bias is not None, eps, skc is not None)
+                    is_stem, bias is not None, eps, skc is not None,
+                    stem_col)
         if use_bn and training:
             ctx.save_for_backward(xc, weight, gamma, beta, y_lin, mean,
                                   rstd, skc)
@@ -147,7 +172,7 @@ class _ConvBNActFn(torch.autograd.Function):
     def backward(ctx, dy):
         C = _C()
         (kh, kw, stride, pad, act_code, use_bn, training, bf16, is_stem,
-         has_bias, eps, has_skip) = ctx.meta
+         has_bias, eps, has_skip, stem_col) = ctx.meta
         dgamma = dbeta = dbias = dskip = None
 
         if use_bn and training:
@@ -183,13 +208,12 @@ class _ConvBNActFn(torch.autograd.Function):
             side = _wgrad_stream(xc.device)
             side.wait_stream(cur)
             with torch.cuda.stream(side):
-                if is_stem and xc.dtype == torch.bfloat16 and kh == 7:
-                    # im2col the 3-channel input to [px][152] rows and run
-                    # the MFMA wgrad as its 1x1 case (~2.4x faster than the
-                    # direct VALU stem_wgrad); column t*3+ci -> dW[co][ci][t]
-                    xcol = C.stem_im2col(xc, kh, stride, pad)
+                if stem_col:
+                    # xc is the SAVED unfolded [px][152] tensor from the
+                    # forward: the wgrad is its 1x1 MFMA case; column
+                    # t*3+ci -> dW[co][ci][t]
                     cout = dpre.shape[1]
-                    dwc = C.wgrad_bf16_fast(xcol, dpre, 1, 1, 1, 0)
+                    dwc = C.wgrad_bf16_fast(xc, dpre, 1, 1, 1, 0)
                     dw = (dwc[:, :kh * kw * 3, 0, 0]
                           .reshape(cout, kh * kw, 3).permute(0, 2, 1)
                           .reshape(cout, 3, kh, kw).contiguous())
@@ -325,13 +349,25 @@ def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
         else:
             scale = torch.ones(cout, device=x.device, dtype=torch.float32)
             shift = bias_f
-        wpk = None if is_stem else _C().pack_weights(weight, False, bf16)
+        if is_stem:
+            wpk = (_C().pack_weights(_stem_col_weight(weight), False, True)
+                   if bf16 else None)
+        else:
+            wpk = _C().pack_weights(weight, False, bf16)
         conv._rthd_infer_cache = (key, wpk, scale, shift)
         cache = conv._rthd_infer_cache
     _, wpk, scale, shift = cache
 
     if is_stem:
         assert skc is None
+        if bf16:
+            # unfold + 1x1 MFMA conv (see _ConvBNActFn): faster than the
+            # direct VALU stem kernel. MUST go through the dispatcher op —
+            # a pybind call here would bake the warmup input's unfolded
+            # tensor into the traced export as a constant.
+            xcol = ops.stem_im2col(xc, kh, stride, pad)
+            return ops.conv_fwd(xcol, wpk, scale, shift, None, 1, 1, 1, 0,
+                                cout, act_code)
         return ops.stem_fwd(xc, weight, scale, shift, stride, pad, act_code)
     return ops.conv_fwd(xc, wpk, scale, shift, skc, kh, kw, stride, pad,
                         cout, act_code)

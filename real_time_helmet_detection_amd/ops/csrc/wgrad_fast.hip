@@ -254,7 +254,21 @@ __global__ void wgrad_reduce_kernel(const float* __restrict__ part,
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        i < n4; i += (int64_t)gridDim.x * blockDim.x) {
     float4 a = {0.f, 0.f, 0.f, 0.f};
-    for (int k = 0; k < K; ++k) {
+    // 4x k-unroll: one outstanding load per thread left this kernel
+    // latency-bound (14 us for ~3 us of traffic)
+    int k = 0;
+    for (; k + 3 < K; k += 4) {
+      float4 v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        v[u] = *reinterpret_cast<const float4*>(
+            &part[(int64_t)(k + u) * N + i * 4]);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        a.x += v[u].x; a.y += v[u].y; a.z += v[u].z; a.w += v[u].w;
+      }
+    }
+    for (; k < K; ++k) {
       const float4 v =
           *reinterpret_cast<const float4*>(&part[(int64_t)k * N + i * 4]);
       a.x += v.x; a.y += v.y; a.z += v.z; a.w += v.w;

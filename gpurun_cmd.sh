@@ -1,7 +1,9 @@
 cd /root/repo
 export TMPDIR=/tmp
 mkdir -p gpurun_out
-timeout 500 python -m pytest tests/test_gpu_kernels.py -x -q -k "wgrad or conv or stem" > gpurun_out/pytest_k.log 2>&1
-echo "P_RC=$?"; tail -1 gpurun_out/pytest_k.log
-timeout 300 python tools/kbench.py wgrad --iters 30 2>/dev/null | grep wgrad
-timeout 300 python bench.py --steps 20 --warmup 8 2>/dev/null | tail -1
+timeout 700 python -m pytest tests -m gpu -q > gpurun_out/pytest_gpu.log 2>&1
+echo "PG_RC=$?"; tail -1 gpurun_out/pytest_gpu.log
+timeout 400 python bench.py --steps 30 --warmup 10 > gpurun_out/bench_train.json 2>/dev/null
+echo "BT_RC=$?"; cat gpurun_out/bench_train.json
+timeout 300 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/bench_infer_b8.json 2>/dev/null
+cat gpurun_out/bench_infer_b8.json

@@ -1,5 +1,8 @@
-cd /root/repo
+cd /tmp
 export TMPDIR=/tmp
-mkdir -p gpurun_out
-timeout 330 python tools/quality_probe.py --steps 2000 --imgs 8 --size 256 > gpurun_out/quality_probe.log 2>&1
-echo "QP_RC=$?"; grep -v Warning gpurun_out/quality_probe.log | tail -10
+mkdir -p /root/repo/gpurun_out
+timeout 240 rocprofv3 --pmc SQ_LDS_BANK_CONFLICT SQ_INSTS_MFMA SQ_INSTS_VALU SQ_WAIT_INST_ANY SQ_WAVE_CYCLES SQ_ACTIVE_INST_ANY -d /tmp/pmcw2 -- python /root/repo/tools/kbench.py wgrad --iters 10 > /root/repo/gpurun_out/pmc2.log 2>&1
+echo "PMC_RC=$?"
+cp -r /tmp/pmcw2 /root/repo/gpurun_out/pmcw2 2>/dev/null
+cd /root/repo
+timeout 240 python tools/kbench.py conv --iters 30 2>/dev/null | grep -i conv

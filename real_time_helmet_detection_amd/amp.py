@@ -30,7 +30,17 @@ def is_autocast_enabled():
 def autocast(enabled=True):
     global _autocast_depth
     if not enabled:
-        yield
+        # torch semantics: a nested disabled region suspends autocast
+        prev = _autocast_depth
+        _autocast_depth = 0
+        try:
+            if torch.cuda.is_available():
+                with torch.autocast(device_type='cuda', enabled=False):
+                    yield
+            else:
+                yield
+        finally:
+            _autocast_depth = prev
         return
     _autocast_depth += 1
     try:

@@ -28,9 +28,12 @@ def init_process_group_from_args(args, rank, world_size):
     backend = args.dist_backend
     if not torch.cuda.is_available() and backend == 'nccl':
         backend = 'gloo'
+    # under torchrun, rendezvous on its MASTER_ADDR/PORT (env://) instead
+    # of the --dist-url default, which may name a different port
+    init_method = 'env://' if launched_from_torchrun() else args.dist_url
     dist.init_process_group(
         backend=backend,
-        init_method=args.dist_url,
+        init_method=init_method,
         world_size=world_size,
         rank=rank,
         timeout=datetime.timedelta(seconds=300),

@@ -20,5 +20,5 @@ if __name__ == '__main__':
         distributed_device_train(args)
     else:
         single_device_evaluate(args)
-    print('%s: Process is Done During %s'
-          % (time.ctime(), str(timedelta(seconds=(time.time() - tictoc)))))
+    print('%s: finished in %s'
+          % (time.ctime(), timedelta(seconds=time.time() - tictoc)))

@@ -48,7 +48,11 @@ def make_parser():
     parser.add_argument('--batch-size', type=int, default=16,
             help='global batch size (split across GPUs)')
     parser.add_argument('--sub-divisions', type=int, default=1,
-            help='optimize every N iterations for gradient accumulation')
+            help='optimize every N iterations for gradient accumulation. '
+                 'NOTE: unlike the reference (which summed micro-batch '
+                 'gradients), the accumulated loss is AVERAGED over the N '
+                 'micro-batches (standard semantics); with N>1 the '
+                 'effective step is 1/N of reference-tuned values')
     parser.add_argument('--start-epoch', type=int, default=0, help='start epoch')
     parser.add_argument('--end-epoch', type=int, default=100, help='end epoch')
     parser.add_argument('--num-workers', type=int, default=8,

@@ -122,7 +122,32 @@ class TrainAugmentor:
         self.affine_scale = affine_scale
         self.multiscale_flag = multiscale_flag
         self.multiscale = list(multiscale)
-        self.rng = np.random.RandomState(seed)
+        self._seed = seed
+        self._rng = None
+        self._worker_seed = None
+
+    @property
+    def rng(self):
+        """Per-worker RandomState: forked DataLoader workers inherit the
+        parent's RNG, so a state created in __init__ would emit IDENTICAL
+        augmentation streams in every worker. Derive the state lazily from
+        torch's per-worker seed (distinct per worker AND per epoch with
+        persistent_workers, via set_epoch -> base_seed reseeding)."""
+        info = None
+        try:
+            import torch.utils.data as tud
+            info = tud.get_worker_info()
+        except Exception:
+            pass
+        wseed = info.seed if info is not None else None
+        if self._rng is None or wseed != self._worker_seed:
+            if wseed is not None:
+                seed = (wseed + (self._seed or 0)) % (2 ** 32)
+            else:
+                seed = self._seed
+            self._rng = np.random.RandomState(seed)
+            self._worker_seed = wseed
+        return self._rng
 
     def __call__(self, img_lst, boxes_lst, labels_lst):
         rng = self.rng

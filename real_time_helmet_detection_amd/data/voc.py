@@ -19,7 +19,6 @@ no-network benchmark configs (BASELINE.json) and CPU tests.
 
 import os
 import time
-import collections
 import xml.etree.ElementTree as ET
 
 import numpy as np
@@ -33,24 +32,33 @@ INDEX2CLASS = {0: 'hat', 1: 'person'}
 CLASS2COLOR = {0: (255, 0, 0), 1: (0, 255, 0)}
 
 
-def parse_voc_xml(node):
-    """Recursive XML -> nested dict (same contract as reference data.py:65-80)."""
-    voc_dict = {}
+def _element_value(node):
+    """XML element -> python value: leaf tags give their stripped text,
+    interior tags give a dict of child values. A tag repeated among its
+    siblings collapses into a list; under <annotation>, 'object' stays a
+    list even when a single <object> is present (the eval driver and
+    boxes_from_voc_dict rely on that — reference data.py:65-80 contract)."""
     children = list(node)
-    if children:
-        def_dic = collections.defaultdict(list)
-        for dc in map(parse_voc_xml, children):
-            for ind, v in dc.items():
-                def_dic[ind].append(v)
-        if node.tag == 'annotation':
-            def_dic['object'] = [def_dic['object']]
-        voc_dict = {node.tag: {ind: v[0] if len(v) == 1 else v
-                               for ind, v in def_dic.items()}}
-    if node.text:
-        text = node.text.strip()
-        if not children:
-            voc_dict[node.tag] = text
-    return voc_dict
+    if not children:
+        return (node.text or '').strip()
+    grouped = {}
+    for child in children:
+        v = _element_value(child)
+        if v == '':  # empty leaf tags carry no information — drop them
+            continue
+        grouped.setdefault(child.tag, []).append(v)
+    out = {}
+    for tag, values in grouped.items():
+        keep_list = len(values) > 1 or (node.tag == 'annotation'
+                                        and tag == 'object')
+        out[tag] = values if keep_list else values[0]
+    return out
+
+
+def parse_voc_xml(node):
+    """XML root -> {tag: nested dict} (same shape the reference produces)."""
+    value = _element_value(node)
+    return {node.tag: value} if value != '' else {}
 
 
 def boxes_from_voc_dict(voc_dict):

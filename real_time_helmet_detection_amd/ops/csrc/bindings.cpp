@@ -73,7 +73,6 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
                                       int64_t act,
                                       c10::optional<torch::Tensor> skip);
 torch::Tensor col_sum(torch::Tensor x);
-torch::Tensor tr16_probe(torch::Tensor in);
 }  // namespace rthd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -102,7 +101,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &rthd::bn_act_fwd, py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());
   m.def("bn_act_bwd", &rthd::bn_act_bwd, py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());
   m.def("col_sum", &rthd::col_sum);
-  m.def("tr16_probe", &rthd::tr16_probe);
 }
 
 // ---------------------------------------------------------------------------

@@ -28,6 +28,19 @@ def _maybe_bf16(x):
     return x
 
 
+def bn_momentum(bn, training):
+    """nn.BatchNorm2d bookkeeping the functional call skips: increment
+    num_batches_tracked on training forwards and resolve momentum=None to
+    the cumulative-moving-average factor 1/num_batches_tracked."""
+    if training and bn.num_batches_tracked is not None:
+        bn.num_batches_tracked.add_(1)
+    if bn.momentum is None:
+        n = (int(bn.num_batches_tracked.item())
+             if bn.num_batches_tracked is not None else 1)
+        return 1.0 / max(n, 1)
+    return bn.momentum
+
+
 def _named_act(y, act):
     """Apply the fused-epilogue activation names on the eager path."""
     if act == 'ReLU':
@@ -58,8 +71,8 @@ def conv_bn_act(x, conv, bn=None, act='Linear', act_module=None,
                  padding=conv.padding)
     if bn is not None:
         y = F.batch_norm(y, bn.running_mean, bn.running_var, bn.weight,
-                         bn.bias, training=training, momentum=bn.momentum,
-                         eps=bn.eps)
+                         bn.bias, training=training,
+                         momentum=bn_momentum(bn, training), eps=bn.eps)
     if skip is not None:
         y = y + skip
     if act_module is not None:

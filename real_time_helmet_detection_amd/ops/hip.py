@@ -398,6 +398,13 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
         if act_module is not None:
             y = act_module(y)
         return y
+    if training:
+        # a training forward mutates BN running stats through raw kernel
+        # writes that never bump buffer._version — drop the inference
+        # caches so a later eval forward re-folds fresh stats
+        conv._rthd_infer_cache = None
+        conv._rthd_fp8_cache = None
+    from .functional import bn_momentum
     y = _ConvBNActFn.apply(
         x, conv.weight,
         conv.bias,
@@ -407,7 +414,7 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
         bn.running_var if use_bn else None,
         skip,
         kh, kw, stride, pad, act_code, use_bn, training,
-        bn.momentum if use_bn else 0.1,
+        bn_momentum(bn, training) if use_bn else 0.1,
         bn.eps if use_bn else 1e-5,
         is_stem)
     if act_module is not None:
@@ -572,5 +579,14 @@ def batched_decode(heatmap, offset, wh, scale_factor, topk, pool_size,
     return boxes, clss, scores
 
 
+_NMS_CAP = 2048  # LDS-resident kernel limit (nms.hip)
+
+
 def nms(boxes, scores, iou_threshold):
+    if boxes.shape[0] > _NMS_CAP:
+        # configs like --topk 1000 with num_stack>=3 and conf_th=0 exceed
+        # the LDS-resident kernel; run the eager O(N^2) suppression instead
+        # of aborting (reference used torchvision.ops.nms with no cap)
+        from .eager import nms as eager_nms
+        return eager_nms(boxes, scores, iou_threshold)
     return _C().nms_fwd(boxes, scores, float(iou_threshold))

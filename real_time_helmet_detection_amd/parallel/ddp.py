@@ -57,6 +57,8 @@ class BucketedDataParallel(nn.Module):
         self.world_size = dist.get_world_size(process_group) \
             if dist.is_initialized() else 1
         self._sync = True
+        self._sync_pass = False
+        self._warned_partial = False
 
         self._device = next(module.parameters()).device
         self._use_comm_stream = self._device.type == 'cuda'
@@ -110,6 +112,7 @@ class BucketedDataParallel(nn.Module):
     def _grad_ready(self, param):
         if not self._sync or self.world_size <= 1:
             return
+        self._sync_pass = True
         bucket = self._param_bucket[param]
         bucket.ready += 1
         if bucket.ready == len(bucket.params):
@@ -156,14 +159,34 @@ class BucketedDataParallel(nn.Module):
             self._sync = prev
 
     def finish_backward(self):
-        """Wait for pending all-reduces and unpack into param.grad."""
-        if self.world_size <= 1 or not self._sync:
+        """Wait for pending all-reduces and unpack into param.grad.
+
+        Every bucket is all-reduced on every sync step, even one whose
+        params received no (or only some) gradients this step: ranks must
+        launch IDENTICAL collective sequences, and a rank whose autograd
+        graph skipped a parameter would otherwise silently desync weights
+        against ranks that did produce that gradient. Unused params
+        contribute zeros (pack_and_reduce packs grad=None as 0)."""
+        if self.world_size <= 1 or not self._sync or not self._sync_pass:
+            # world 1, inside no_sync, or no synced backward ran since the
+            # last finish (e.g. finish_backward after a no_sync-only pass):
+            # nothing to communicate
             self._reset_ready()
             return
         launched = False
         for b in self.buckets:
-            if b.ready == len(b.params) and b.work is None:
-                # hook raced/skipped (shouldn't happen) — launch now
+            if b.work is None:
+                # bucket never became fully ready in backward (some params
+                # got no grad this step) — flush it now so the collective
+                # sequence matches the other ranks
+                if 0 < b.ready < len(b.params) and not self._warned_partial:
+                    self._warned_partial = True
+                    import warnings
+                    warnings.warn(
+                        'BucketedDataParallel: a gradient bucket was only '
+                        'partially ready at finish_backward (%d/%d grads); '
+                        'missing grads sync as zeros' %
+                        (b.ready, len(b.params)))
                 self._launch_bucket(b)
             if b.work is not None:
                 b.work.wait()
@@ -183,12 +206,18 @@ class BucketedDataParallel(nn.Module):
         for b in self.buckets:
             for p, off in zip(b.params, b.offsets):
                 if p.grad is None:
+                    # this rank produced no grad but another may have —
+                    # materialize the averaged result so optimizer.step
+                    # applies the same update on every rank
+                    p.grad = b.flat[off:off + p.numel()].to(p.dtype) \
+                        .view_as(p).clone()
                     continue
                 p.grad.detach().reshape(-1).copy_(
                     b.flat[off:off + p.numel()].to(p.grad.dtype),
                     non_blocking=True)
 
     def _reset_ready(self):
+        self._sync_pass = False
         for b in self.buckets:
             b.ready = 0
 

@@ -18,8 +18,12 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, 'real_time_helmet_detection_amd', 'ops', 'csrc')
 
-sources = sorted(glob.glob(os.path.join(CSRC, '*.cpp')) +
-                 glob.glob(os.path.join(CSRC, '*.hip')))
+# exclude hipify build artifacts (*_hip.hip) that torch's build may drop
+# next to the real sources — they are gitignored copies, not sources
+sources = sorted(p for p in
+                 glob.glob(os.path.join(CSRC, '*.cpp')) +
+                 glob.glob(os.path.join(CSRC, '*.hip'))
+                 if not p.endswith('_hip.hip'))
 
 setup(
     name='real_time_helmet_detection_amd',

@@ -111,6 +111,73 @@ def test_no_sync_skips_communication(tmp_path):
         torch.testing.assert_close(g, w)
 
 
+class _TwoHead(torch.nn.Module):
+    """Model whose second head gets no gradient when use_b=False."""
+
+    def __init__(self):
+        super().__init__()
+        self.a = torch.nn.Linear(8, 8)
+        self.b = torch.nn.Linear(8, 8)
+
+    def forward(self, x, use_b=True):
+        y = self.a(x)
+        if use_b:
+            y = y + self.b(x)
+        return y
+
+
+def _worker_unused_param(rank, world, rdv_file, result_dir):
+    from real_time_helmet_detection_amd.parallel.ddp import \
+        BucketedDataParallel
+    dist.init_process_group('gloo', init_method=f'file://{rdv_file}',
+                            world_size=world, rank=rank)
+    try:
+        torch.manual_seed(0)
+        net = _TwoHead()
+        ddp = BucketedDataParallel(net, bucket_cap_mb=64.0)
+        g = torch.Generator().manual_seed(100 + rank)
+        x = torch.randn(4, 8, generator=g)
+        # rank 0 skips head b entirely: its bucket never becomes fully
+        # ready in backward and must be flushed by finish_backward (zeros
+        # for the missing grads) so the collective sequences match
+        ddp(x, use_b=(rank != 0)).sum().backward()
+        ddp.finish_backward()
+        torch.save({n: p.grad for n, p in net.named_parameters()},
+                   os.path.join(result_dir, 'grads_rank%d.pt' % rank))
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_partially_ready_bucket_flushed(tmp_path):
+    """A rank whose autograd graph skips params must neither hang nor
+    desync: finish_backward flushes the partial bucket with zeros and the
+    averaged gradient is identical (and materialized) on both ranks."""
+    world = 2
+    rdv = tmp_path / 'rdv_unused'
+    mp.spawn(_worker_unused_param, nprocs=world,
+             args=(world, str(rdv), str(tmp_path)), join=True)
+    g0 = torch.load(tmp_path / 'grads_rank0.pt', weights_only=False)
+    g1 = torch.load(tmp_path / 'grads_rank1.pt', weights_only=False)
+
+    # oracle: rank0 grads with b unused (b-grads = 0), rank1 with b used
+    def local_grads(rank):
+        torch.manual_seed(0)
+        net = _TwoHead()
+        g = torch.Generator().manual_seed(100 + rank)
+        x = torch.randn(4, 8, generator=g)
+        net(x, use_b=(rank != 0)).sum().backward()
+        return {n: (p.grad if p.grad is not None else torch.zeros_like(p))
+                for n, p in net.named_parameters()}
+
+    l0, l1 = local_grads(0), local_grads(1)
+    for name in g0:
+        want = (l0[name] + l1[name]) / world
+        assert g0[name] is not None, f'{name}: grad not materialized'
+        torch.testing.assert_close(g0[name], want, rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(g1[name], want, rtol=1e-5, atol=1e-6)
+
+
 def test_bucket_partitioning():
     from real_time_helmet_detection_amd.parallel.ddp import \
         BucketedDataParallel

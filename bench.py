@@ -128,7 +128,10 @@ def main():
 
     graph_active = False
     if args.mode == 'train':
-        want_graph = use_cuda and world == 1 and not args.no_train_graph
+        # N>1 captures the bucketed RCCL all-reduce into the graph too
+        # (warmup establishes the comm clique first); any capture failure
+        # falls back to eager stepping
+        want_graph = use_cuda and not args.no_train_graph
         try:
             opt = torch.optim.Adam(net.parameters(), lr=5e-4, fused=True,
                                    capturable=want_graph)

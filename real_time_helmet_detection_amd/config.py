@@ -171,6 +171,14 @@ def make_parser():
             help='NHWC tensor layout (the HIP kernels are NHWC-native)')
     parser.add_argument('--no-channels-last', dest='channels_last',
             action='store_false')
+    parser.add_argument('--train-graph', dest='train_graph',
+            action='store_true', default=True,
+            help='hipGraph-capture the training step (one graph launch '
+                 'per iteration instead of ~150 kernel launches); '
+                 'auto-disabled for --sub-divisions > 1 and on any '
+                 'capture failure')
+    parser.add_argument('--no-train-graph', dest='train_graph',
+            action='store_false')
     parser.add_argument('--synthetic', action='store_true', default=False,
             help='use synthetic VOC2028-shaped data (no dataset on disk)')
     parser.add_argument('--synthetic-size', type=int, default=512,

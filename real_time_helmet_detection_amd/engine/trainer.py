@@ -108,24 +108,52 @@ def compute_stack_losses(outputs, loss_calculator, gt_heatmap, gt_offset,
     """Deep supervision: sum the loss over every stack's prediction.
 
     outputs: (B, S, num_cls+4, h, w) raw logits. Sigmoid on the heatmap
-    channels (and on offset/size when normalized_coord) happens here, outside
-    the network (reference train.py:104-120) — in fp32.
+    channels (and on offset/size when normalized_coord) is applied on the
+    way into the loss, outside the network (reference train.py:104-120), in
+    fp32. On GPU the whole thing — sigmoid, upcast, all stacks — is ONE
+    fused kernel pair (ops.hip.centernet_losses_logits).
+
+    Returns (total_loss, last-stack heatmap LOGITS) — apply sigmoid before
+    visualizing the heatmap.
     """
+    from ..ops import _backend
+    if outputs.is_cuda and not _backend.eager_gpu_override():
+        from ..ops import hip
+        losses = hip.centernet_losses_logits(
+            outputs, gt_heatmap, gt_offset, gt_size, gt_mask,
+            loss_calculator.focal_alpha, loss_calculator.focal_beta,
+            normalized_coord)
+        total = loss_calculator.accumulate_stack_losses(losses)
+        return total, outputs[:, -1, :num_cls]
+
     total = None
     last_heatmap = None
     for output in outputs.split(1, dim=1):
         output = output.squeeze(1).float()
-        pred_heatmap, pred_offset, pred_size = output.split(
+        hm_logits, pred_offset, pred_size = output.split(
             [num_cls, 2, 2], dim=1)
-        pred_heatmap = torch.sigmoid(pred_heatmap)
+        pred_heatmap = torch.sigmoid(hm_logits)
         if normalized_coord:
             pred_offset = torch.sigmoid(pred_offset)
             pred_size = torch.sigmoid(pred_size)
         loss = loss_calculator(pred_heatmap, pred_offset, pred_size,
                                gt_heatmap, gt_offset, gt_size, gt_mask)
         total = loss if total is None else total + loss
-        last_heatmap = pred_heatmap
+        last_heatmap = hm_logits
     return total, last_heatmap
+
+
+def _want_train_graph(args, device):
+    """hipGraph capture of the full training step: GPU + HIP engine +
+    no gradient accumulation (micro-steps change the per-iteration op
+    sequence)."""
+    device = torch.device(device) if not isinstance(device, torch.device) \
+        else device
+    return (getattr(args, 'train_graph', True)
+            and device.type == 'cuda'
+            and args.train_flag
+            and getattr(args, 'sub_divisions', 1) == 1
+            and os.environ.get('RTHD_EAGER_GPU') != '1')
 
 
 def train_step(dataloader, network, loss_calculator, optimizer, scheduler,
@@ -138,6 +166,14 @@ def train_step(dataloader, network, loss_calculator, optimizer, scheduler,
     n_batches = len(dataloader)
 
     is_bucketed = isinstance(network, BucketedDataParallel)
+
+    graphed = getattr(network, '_rthd_graphed', None)
+    if graphed is None and _want_train_graph(args, dev):
+        from .graphed import GraphedTrainStep
+        graphed = GraphedTrainStep(network, loss_calculator, optimizer,
+                                   args.num_cls, args.normalized_coord,
+                                   amp_on=(scaler is not None))
+        network._rthd_graphed = graphed
 
     tictoc = time.time()
     for iteration, (image, gt_heatmap, gt_offset, gt_size, gt_mask,
@@ -155,6 +191,22 @@ def train_step(dataloader, network, loss_calculator, optimizer, scheduler,
         gt_offset = gt_offset.to(dev, non_blocking=True)
         gt_size = gt_size.to(dev, non_blocking=True)
         gt_mask = gt_mask.to(dev, non_blocking=True)
+
+        if graphed is not None and graphed.enabled:
+            # whole iteration (fwd + fused loss + bwd + allreduce + Adam)
+            # as one hipGraph replay; falls through to eager on shape
+            # overflow or capture failure
+            hm_logits = graphed.step(image, gt_heatmap, gt_offset, gt_size,
+                                     gt_mask)
+            if hm_logits is not None:
+                pred_heatmap = hm_logits
+                time_logger['forward'].update(time.time() - tictoc)
+                if (iteration % args.print_interval == 0) and (device == 0):
+                    _print_train_log(time_logger, loss_calculator, epoch,
+                                     iteration, n_batches, args, image,
+                                     pred_heatmap, gt_heatmap)
+                tictoc = time.time()
+                continue
 
         with amp.autocast(enabled=scaler is not None):
             outputs = network(image)
@@ -188,27 +240,36 @@ def train_step(dataloader, network, loss_calculator, optimizer, scheduler,
         time_logger['backward'].update(time.time() - tictoc)
 
         if (iteration % args.print_interval == 0) and (device == 0):
-            loss_log = loss_calculator.get_log()
-            _log = '%s: Epoch [%2d/%2d]' % (time.ctime(), epoch,
-                                            args.end_epoch)
-            _log += ', Iteration [%4d/%4d]' % (iteration, n_batches)
-            _log += ', Loss [%s]' % loss_log
-            _log += ', Time(ms) [data: %6.2f' % (time_logger['data'].avg * 1e3)
-            _log += ', forward: %6.2f' % (time_logger['forward'].avg * 1e3)
-            _log += ', backward: %6.2f' % (time_logger['backward'].avg * 1e3)
-            _log += ', loss: %6.2f]' % (time_logger['loss'].avg * 1e3)
-            print(_log)
-
-            log_dir = os.path.join(args.save_path, 'training_log')
-            if os.path.isdir(log_dir):
-                blend_heatmap(image[0], pred_heatmap[0],
-                              args.pretrained).save(
-                    os.path.join(log_dir, 'training_pred.png'))
-                blend_heatmap(image[0], gt_heatmap[0],
-                              args.pretrained).save(
-                    os.path.join(log_dir, 'training_gt.png'))
+            _print_train_log(time_logger, loss_calculator, epoch, iteration,
+                             n_batches, args, image, pred_heatmap,
+                             gt_heatmap)
 
         tictoc = time.time()
+
+
+def _print_train_log(time_logger, loss_calculator, epoch, iteration,
+                     n_batches, args, image, pred_heatmap, gt_heatmap):
+    """Rank-0 progress line + heatmap-blend PNGs (reference
+    train.py:141-158). pred_heatmap holds LOGITS (sigmoid applied here,
+    only when actually rendering)."""
+    loss_log = loss_calculator.get_log()
+    _log = '%s: Epoch [%2d/%2d]' % (time.ctime(), epoch, args.end_epoch)
+    _log += ', Iteration [%4d/%4d]' % (iteration, n_batches)
+    _log += ', Loss [%s]' % loss_log
+    _log += ', Time(ms) [data: %6.2f' % (time_logger['data'].avg * 1e3)
+    _log += ', forward: %6.2f' % (time_logger['forward'].avg * 1e3)
+    _log += ', backward: %6.2f' % (time_logger['backward'].avg * 1e3)
+    _log += ', loss: %6.2f]' % (time_logger['loss'].avg * 1e3)
+    print(_log)
+
+    log_dir = os.path.join(args.save_path, 'training_log')
+    if os.path.isdir(log_dir):
+        blend_heatmap(image[0], torch.sigmoid(pred_heatmap[0].float()),
+                      args.pretrained).save(
+            os.path.join(log_dir, 'training_pred.png'))
+        blend_heatmap(image[0], gt_heatmap[0],
+                      args.pretrained).save(
+            os.path.join(log_dir, 'training_gt.png'))
 
 
 def load_network(args, device):
@@ -240,7 +301,8 @@ def load_network(args, device):
     if args.train_flag:
         optimizer, scheduler = get_optimizer(
             network=network, lr=args.lr, lr_milestone=args.lr_milestone,
-            lr_gamma=args.lr_gamma, algo=args.optim)
+            lr_gamma=args.lr_gamma, algo=args.optim,
+            capturable=_want_train_graph(args, device))
         loss_calculator = LossCalculator(
             hm_weight=args.hm_weight,
             offset_weight=args.offset_weight,

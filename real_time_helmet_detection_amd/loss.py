@@ -87,6 +87,23 @@ class LossCalculator(nn.Module):
                               size_loss.detach(), total_loss.detach()))
         return total_loss
 
+    def accumulate_stack_losses(self, losses):
+        """Weight + log a [S, 3] per-stack (hm, offset, size) loss tensor
+        (the fused all-stacks GPU path) and return the summed total."""
+        w = getattr(self, '_weight_vec', None)
+        if w is None or w.device != losses.device:
+            w = torch.tensor([self.hm_weight, self.offset_weight,
+                              self.size_weight], device=losses.device,
+                             dtype=losses.dtype)
+            self._weight_vec = w
+        per_stack = losses @ w          # [S] weighted totals
+        total = per_stack.sum()
+        det, ps_det = losses.detach(), per_stack.detach()
+        for i in range(losses.shape[0]):
+            self._pending.append((det[i, 0], det[i, 1], det[i, 2],
+                                  ps_det[i]))
+        return total
+
     def flush_log(self):
         """Materialize pending device scalars into ``.log`` (one sync)."""
         if not self._pending:

@@ -27,6 +27,14 @@ std::vector<torch::Tensor> centernet_loss_bwd(
     torch::Tensor ghm, torch::Tensor goff, torch::Tensor gsize,
     torch::Tensor mask, torch::Tensor sums, torch::Tensor gout,
     double alpha, double beta);
+std::vector<torch::Tensor> centernet_loss_fused_fwd(
+    torch::Tensor out, torch::Tensor ghm, torch::Tensor goff,
+    torch::Tensor gsize, torch::Tensor mask, double alpha, double beta,
+    bool sig_os);
+torch::Tensor centernet_loss_fused_bwd(
+    torch::Tensor out, torch::Tensor ghm, torch::Tensor goff,
+    torch::Tensor gsize, torch::Tensor mask, torch::Tensor sums,
+    torch::Tensor glosses, double alpha, double beta, bool sig_os);
 
 std::vector<torch::Tensor> decode_fwd(torch::Tensor hm, torch::Tensor off,
                                       torch::Tensor wh, int64_t scale_factor,
@@ -86,6 +94,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upsample2x_bwd", &rthd::upsample2x_bwd);
   m.def("centernet_loss_fwd", &rthd::centernet_loss_fwd);
   m.def("centernet_loss_bwd", &rthd::centernet_loss_bwd);
+  m.def("centernet_loss_fused_fwd", &rthd::centernet_loss_fused_fwd);
+  m.def("centernet_loss_fused_bwd", &rthd::centernet_loss_fused_bwd);
   m.def("decode_fwd", &rthd::decode_fwd);
   m.def("nms_fwd", &rthd::nms_fwd);
   m.def("pack_weights", &rthd::pack_weights);

@@ -60,6 +60,18 @@ def _bf16_mode(x):
     return x.dtype == torch.bfloat16 or _amp.is_autocast_enabled()
 
 
+# Monotone stamp bumped on every training-mode conv forward (and on every
+# hipGraph replay of a captured training step, where no python runs): BN
+# running stats are updated by raw kernel writes that never bump
+# tensor._version, so the folded-BN inference caches key on this stamp to
+# stay fresh across train->eval->train transitions.
+_train_stamp = [0]
+
+
+def bump_train_stamp():
+    _train_stamp[0] += 1
+
+
 # per-(device, n) constant epilogue vectors: the training path needs
 # ones/zeros per conv call — allocating them fresh was ~130 tiny kernel
 # launches per train step
@@ -286,7 +298,7 @@ def _conv_infer_fp8(x, conv, bn, act_code):
     stride, pad = conv.stride[0], conv.padding[0]
     cout = conv.weight.shape[0]
     cache = getattr(conv, '_rthd_fp8_cache', None)
-    ver = conv.weight._version
+    ver = (conv.weight._version, _train_stamp[0])
     if cache is None or cache[0] != ver:
         w = conv.weight.detach().float()
         sw = w.abs().amax(dim=(1, 2, 3)).clamp(min=1e-8) / 240.0
@@ -329,7 +341,7 @@ def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
     weight = conv.weight
     cout = weight.shape[0]
 
-    key = (bf16, weight._version,
+    key = (bf16, _train_stamp[0], weight._version,
            conv.bias._version if conv.bias is not None else -1,
            bn.weight._version if bn is not None else -1,
            bn.bias._version if bn is not None else -1,
@@ -400,10 +412,9 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
         return y
     if training:
         # a training forward mutates BN running stats through raw kernel
-        # writes that never bump buffer._version — drop the inference
-        # caches so a later eval forward re-folds fresh stats
-        conv._rthd_infer_cache = None
-        conv._rthd_fp8_cache = None
+        # writes that never bump buffer._version — bump the stamp the
+        # inference caches key on, so a later eval forward re-folds
+        bump_train_stamp()
     from .functional import bn_momentum
     y = _ConvBNActFn.apply(
         x, conv.weight,
@@ -567,6 +578,41 @@ def centernet_losses(phm, poff, psize, ghm, goff, gsize, mask,
     return _CenterNetLossFn.apply(phm.float(), poff.float(), psize.float(),
                                   ghm, goff, gsize, mask,
                                   float(focal_alpha), float(focal_beta))
+
+
+class _CenterNetLossLogitsFn(torch.autograd.Function):
+    """All-stacks fused form: raw (B,S,C+4,h,w) LOGITS in, [S,3] losses out.
+
+    The heatmap sigmoid (reference train.py:107-111, applied outside the
+    network) and the fp32 upcast live inside the kernel; backward produces
+    d/d_logit in the logits' dtype. One kernel pair per STEP instead of
+    ~10 launches per stack."""
+
+    @staticmethod
+    def forward(ctx, out, ghm, goff, gsize, mask, alpha, beta, sig_os):
+        out_c = out.contiguous()
+        losses, sums = _C().centernet_loss_fused_fwd(
+            out_c, ghm, goff, gsize, mask, alpha, beta, sig_os)
+        ctx.save_for_backward(out_c, ghm, goff, gsize, mask, sums)
+        ctx.meta = (alpha, beta, sig_os)
+        return losses
+
+    @staticmethod
+    def backward(ctx, glosses):
+        out_c, ghm, goff, gsize, mask, sums = ctx.saved_tensors
+        alpha, beta, sig_os = ctx.meta
+        dout = _C().centernet_loss_fused_bwd(
+            out_c, ghm, goff, gsize, mask, sums, glosses.contiguous(),
+            alpha, beta, sig_os)
+        return dout, None, None, None, None, None, None, None
+
+
+def centernet_losses_logits(out, ghm, goff, gsize, mask, focal_alpha,
+                            focal_beta, sigmoid_offsize=False):
+    return _CenterNetLossLogitsFn.apply(out, ghm, goff, gsize, mask,
+                                        float(focal_alpha),
+                                        float(focal_beta),
+                                        bool(sigmoid_offsize))
 
 
 # ---------------------------------------------------------------- decode ---

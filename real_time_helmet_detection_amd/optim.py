@@ -8,14 +8,19 @@ experiments (same signature either way).
 import torch.optim as optim
 
 
-def get_optimizer(network, lr, lr_milestone, lr_gamma, algo='Adam'):
+def get_optimizer(network, lr, lr_milestone, lr_gamma, algo='Adam',
+                  capturable=False):
+    """capturable=True keeps the fused-Adam step state on device so the
+    step can be recorded into a hipGraph (engine/graphed.py); only valid
+    for CUDA parameters."""
     algo = (algo or 'Adam').lower()
     params = network.parameters()
     if algo == 'adam':
         # fused Adam: one kernel per dtype group instead of ~200 small
         # elementwise launches per step (measured ~1 ms/step)
         try:
-            optimizer = optim.Adam(params, lr=lr, fused=True)
+            optimizer = optim.Adam(params, lr=lr, fused=True,
+                                   capturable=capturable)
         except (RuntimeError, TypeError, ValueError):
             optimizer = optim.Adam(params, lr=lr, foreach=True)
     elif algo == 'adamw':

@@ -164,6 +164,88 @@ def test_centernet_loss_fwd_bwd():
     assert rel_err(dpsize, psize_g.grad) < 1e-5
 
 
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('sig_os', [False, True])
+def test_centernet_loss_fused_logits(dtype, sig_os):
+    """All-stacks fused logits kernel vs the eager per-stack sigmoid path."""
+    torch.manual_seed(11)
+    B, S, C, h, w = 3, 2, 2, 32, 32
+    out = torch.randn(B, S, C + 4, h, w) * 2
+    if dtype == torch.bfloat16:
+        out = out.to(dtype).float()  # oracle on the rounded values
+    ghm = torch.rand(B, C, h, w)
+    goff = torch.rand(B, 2, h, w)
+    gsize = torch.rand(B, 2, h, w) * 10
+    mask = (torch.rand(B, 1, h, w) > 0.95).float()
+
+    # eager oracle: per-stack sigmoid outside, fp32
+    from real_time_helmet_detection_amd.ops import eager
+    want = []
+    out_g = out.clone().requires_grad_(True)
+    total = 0
+    wvec = (1.0, 2.0, 0.5)
+    for s in range(S):
+        o = out_g[:, s]
+        phm = torch.sigmoid(o[:, :C])
+        poff, psize = o[:, C:C + 2], o[:, C + 2:]
+        if sig_os:
+            poff, psize = torch.sigmoid(poff), torch.sigmoid(psize)
+        hm_l, off_l, sz_l = eager.centernet_losses(
+            phm, poff, psize, ghm, goff, gsize, mask, 2.0, 4.0)
+        want.append((hm_l, off_l, sz_l))
+        total = total + wvec[0] * hm_l + wvec[1] * off_l + wvec[2] * sz_l
+    total.backward()
+
+    from real_time_helmet_detection_amd.ops import hip
+    out_d = out.to('cuda', dtype).requires_grad_(True)
+    gts = [t.cuda() for t in (ghm, goff, gsize, mask)]
+    losses = hip.centernet_losses_logits(out_d, *gts, 2.0, 4.0, sig_os)
+    assert losses.shape == (S, 3)
+    tol = 2e-3 if dtype == torch.bfloat16 else 1e-4
+    for s in range(S):
+        for i in range(3):
+            w_ = want[s][i].item()
+            assert abs(losses[s, i].item() - w_) < tol * max(1.0, abs(w_)), \
+                (s, i, losses[s, i].item(), w_)
+
+    wt = torch.tensor(wvec, device='cuda')
+    (losses * wt).sum().backward()
+    gtol = 0.03 if dtype == torch.bfloat16 else 1e-3
+    assert rel_err(out_d.grad, out_g.grad) < gtol
+
+
+def test_compute_stack_losses_gpu_matches_cpu():
+    """The trainer's fused GPU loss path vs its eager CPU path."""
+    from real_time_helmet_detection_amd.engine.trainer import \
+        compute_stack_losses
+    from real_time_helmet_detection_amd.loss import LossCalculator
+    torch.manual_seed(12)
+    B, S, C, h, w = 2, 2, 2, 32, 32
+    out = torch.randn(B, S, C + 4, h, w)
+    ghm = torch.rand(B, C, h, w)
+    goff = torch.rand(B, 2, h, w)
+    gsize = torch.rand(B, 2, h, w) * 10
+    mask = (torch.rand(B, 1, h, w) > 0.95).float()
+
+    calc_cpu = LossCalculator()
+    want, _ = compute_stack_losses(out, calc_cpu, ghm, goff, gsize, mask,
+                                   C, False)
+    calc_gpu = LossCalculator().cuda()
+    got, hm_logits = compute_stack_losses(
+        out.cuda(), calc_gpu, ghm.cuda(), goff.cuda(), gsize.cuda(),
+        mask.cuda(), C, False)
+    assert abs(got.item() - want.item()) < 1e-3 * max(1.0, abs(want.item()))
+    # returned heatmap is the last stack's LOGITS
+    torch.testing.assert_close(hm_logits.cpu(), out[:, -1, :C])
+    # the per-stack log entries match too
+    calc_cpu.flush_log()
+    calc_gpu.flush_log()
+    for k in calc_cpu.LOG_KEYS:
+        assert len(calc_gpu.log[k]) == len(calc_cpu.log[k]) == S
+        for a, b in zip(calc_gpu.log[k], calc_cpu.log[k]):
+            assert abs(a - b) < 1e-3 * max(1.0, abs(b))
+
+
 # ----------------------------------------------------------------- decode --
 
 def test_decode_matches_eager():

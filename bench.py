@@ -82,7 +82,11 @@ def main():
     local_rank = int(os.environ.get('LOCAL_RANK', '0'))
 
     use_cuda = torch.cuda.is_available()
-    device = torch.device('cuda', local_rank) if use_cuda \
+    # local_rank may exceed the visible device count when de-risking the
+    # RCCL path with multiple ranks sharing one GPU (2 ranks on a 1-GPU
+    # box); fold it into the available devices
+    dev_idx = (local_rank % torch.cuda.device_count()) if use_cuda else 0
+    device = torch.device('cuda', dev_idx) if use_cuda \
         else torch.device('cpu')
     if use_cuda:
         torch.cuda.set_device(device)

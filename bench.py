@@ -168,7 +168,10 @@ def main():
                 graphs = []
                 for b in range(len(batches)):
                     g = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(g):
+                    # thread_local: the RCCL watchdog thread's HIP calls
+                    # must not invalidate an N>1 capture
+                    with torch.cuda.graph(
+                            g, capture_error_mode='thread_local'):
                         eager_step(b)
                     graphs.append(g)
                 calc._pending.clear()

@@ -1,13 +1,17 @@
+set -x
 cd /root/repo
-export TMPDIR=/tmp
-mkdir -p gpurun_out
-timeout 300 python main.py --train-flag --synthetic --synthetic-size 16 --amp \
-  --batch-size 8 --end-epoch 1 --num-workers 2 --imsize 256 \
-  --num-stack 1 --hourglass-inch 64 --print-interval 1 \
-  --save-path gpurun_out/cli_run > gpurun_out/cli_train.log 2>&1
-echo "TRAIN_RC=$?"; tail -2 gpurun_out/cli_train.log
-ls gpurun_out/cli_run/ | head -5
-timeout 200 python main.py --synthetic --synthetic-size 4 --num-workers 0 \
-  --imsize 256 --save-path gpurun_out/cli_run \
-  --model-load gpurun_out/cli_run/check_point_1.pth > gpurun_out/cli_eval.log 2>&1
-echo "EVAL_RC=$?"; tail -2 gpurun_out/cli_eval.log
+python -m pytest tests -m gpu -x -q > gpurun_out/r2_pytest.log 2>&1
+echo "pytest rc=$?"
+timeout 240 python bench.py --steps 30 --warmup 5 > gpurun_out/r2_bench1.json 2>&1
+echo "bench rc=$?"
+timeout 300 python main.py --train-flag --synthetic --synthetic-size 512 --batch-size 16 --amp --end-epoch 2 --print-interval 8 --num-workers 8 --save-path /tmp/w_graph > gpurun_out/r2_cli_graph.log 2>&1
+echo "cli_graph rc=$?"
+timeout 300 python main.py --train-flag --synthetic --synthetic-size 512 --batch-size 16 --amp --end-epoch 2 --print-interval 8 --num-workers 8 --no-train-graph --save-path /tmp/w_eager > gpurun_out/r2_cli_eager.log 2>&1
+echo "cli_eager rc=$?"
+timeout 240 python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 --master-addr 127.0.0.1 --master-port 29511 bench.py --gpus 2 --steps 10 --warmup 3 --batch-size 8 > gpurun_out/r2_rccl2.log 2>&1
+echo "rccl2 rc=$?"
+tail -3 gpurun_out/r2_pytest.log
+tail -2 gpurun_out/r2_bench1.json
+tail -4 gpurun_out/r2_cli_graph.log
+tail -4 gpurun_out/r2_cli_eager.log
+tail -6 gpurun_out/r2_rccl2.log

@@ -91,7 +91,12 @@ class GraphedTrainStep:
         torch.cuda.synchronize()
         n_pending_warm = len(self.calc._pending)
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        # thread_local capture mode: the DataLoader pin-memory thread (and
+        # the RCCL watchdog) make HIP calls concurrently with the capture;
+        # the default 'global' mode invalidates the capture on ANY
+        # thread's unsafe call (measured: hipErrorStreamCaptureInvalidated
+        # the moment a loader is attached)
+        with torch.cuda.graph(g, capture_error_mode='thread_local'):
             losses, per_stack, hm_logits = self._eager_body(*statics)
         # warmup entries are real updates (keep); capture-time entries
         # reference graph-owned memory that every replay overwrites (drop)
@@ -127,6 +132,13 @@ class GraphedTrainStep:
                 print('rthd: train-graph capture failed (%s); '
                       'eager stepping' % exc)
                 self.enabled = False
+                # a failed capture can leave the stream poisoned; drain
+                # it so the eager fallback starts clean
+                try:
+                    torch.cuda.synchronize()
+                except Exception:
+                    pass
+                self.opt.zero_grad(set_to_none=True)
                 return None
             self.graphs[key] = entry
             # the capture already consumed this batch (warmup+capture)

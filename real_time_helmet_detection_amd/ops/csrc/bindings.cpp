@@ -55,6 +55,12 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                        c10::optional<torch::Tensor> skip,
                        int64_t KH, int64_t KW, int64_t stride, int64_t pad,
                        int64_t Cout, int64_t act);
+torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
+                             torch::Tensor scale, torch::Tensor shift,
+                             c10::optional<torch::Tensor> skip,
+                             int64_t KH, int64_t KW, int64_t stride,
+                             int64_t pad, int64_t Cout, int64_t act,
+                             int64_t splitk);
 torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
                        torch::Tensor scale, torch::Tensor shift,
                        int64_t stride, int64_t pad, int64_t act);
@@ -102,6 +108,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_weights_fp8", &rthd::pack_weights_fp8);
   m.def("conv_fwd_fp8", &rthd::conv_fwd_fp8);
   m.def("conv_fwd", &rthd::conv_fwd);
+  m.def("conv_fwd_small", &rthd::conv_fwd_small);
   m.def("stem_fwd", &rthd::stem_fwd);
   m.def("stem_wgrad", &rthd::stem_wgrad);
   m.def("stem_im2col", &rthd::stem_im2col);

@@ -20,6 +20,10 @@
 // stem, which has its own kernel in stem.hip).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+
+#include <mutex>
+#include <unordered_map>
+
 #include "common.h"
 
 namespace rthd {
@@ -446,6 +450,70 @@ void conv_fwd_f32_kernel(const float* __restrict__ x,
   }
 }
 
+// ---------------------------- autotune cache --------------------------------
+// cudnn.benchmark analog (SURVEY.md §2.3 last row): per (shape, dtype) the
+// first eligible call MEASURES the 128x128 kernel against the 64x64
+// (+split-K) variants on the current stream and caches the winner; under
+// stream capture (hipGraph) or RTHD_NO_AUTOTUNE an unseen shape takes the
+// fill-based heuristic instead (no timing APIs are capture-legal).
+
+torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
+                             torch::Tensor scale, torch::Tensor shift,
+                             c10::optional<torch::Tensor> skip,
+                             int64_t KH, int64_t KW, int64_t stride,
+                             int64_t pad, int64_t Cout, int64_t act,
+                             int64_t splitk);
+
+namespace {
+
+struct ConvChoice { int small; int splitk; };
+
+std::mutex g_conv_tune_mu;
+std::unordered_map<uint64_t, ConvChoice> g_conv_tune;
+
+uint64_t conv_key(const ConvGeo& g) {
+  uint64_t h = 1469598103934665603ull;
+  auto mix = [&h](uint64_t v) {
+    h ^= v + 0x9e3779b97f4a7c15ull + (h << 6) + (h >> 2);
+  };
+  mix(g.M); mix(g.Cin); mix(g.Cout); mix(g.KH); mix(g.KW); mix(g.stride);
+  return h;
+}
+
+bool stream_capturing(hipStream_t s) {
+  hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+  if (hipStreamIsCapturing(s, &st) != hipSuccess) {
+    (void)hipGetLastError();
+    return true;  // be conservative: never time inside a capture
+  }
+  return st != hipStreamCaptureStatusNone;
+}
+
+// median-of-3 time of fn() in usec on stream s (fn must enqueue its work
+// on s); synchronizes the stream
+template <typename F>
+float time_usec(F&& fn, hipStream_t s) {
+  hipEvent_t e0, e1;
+  (void)hipEventCreate(&e0);
+  (void)hipEventCreate(&e1);
+  fn();  // warm (code paths, allocator)
+  float best = 1e30f;
+  for (int r = 0; r < 3; ++r) {
+    (void)hipEventRecord(e0, s);
+    fn();
+    (void)hipEventRecord(e1, s);
+    (void)hipEventSynchronize(e1);
+    float ms = 0.f;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    best = std::min(best, ms * 1000.f);
+  }
+  (void)hipEventDestroy(e0);
+  (void)hipEventDestroy(e1);
+  return best;
+}
+
+}  // namespace
+
 // host wrapper; x NCHW-logical channels_last; wpk from pack_weights.
 torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                        torch::Tensor scale, torch::Tensor shift,
@@ -491,6 +559,68 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
     TORCH_CHECK(g.Cin % 8 == 0,
                 "bf16 conv requires Cin % 8 == 0 (glds staging); "
                 "Cin=", g.Cin, " runs the f32 path");
+
+    // -------- per-shape variant selection (autotune cache) --------
+    const int big_blocks = (int)cdiv(g.M, 128) * (g.Coutp / 128);
+    if (big_blocks < 512) {  // big kernel can't fill 256 CUs x 2 waves
+      ConvChoice ch{0, 1};
+      bool have = false;
+      const uint64_t key = conv_key(g);
+      {
+        std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+        auto it = g_conv_tune.find(key);
+        if (it != g_conv_tune.end()) { ch = it->second; have = true; }
+      }
+      if (!have) {
+        // candidate split-K factors for the 64x64 kernel: enough z-blocks
+        // to reach ~1024 blocks, chunk >= 2 K-steps
+        const int nsteps = (int)(KH * KW) * (g.Cinp / 32);
+        const int base64 = (int)cdiv(g.M, 64) * (int)cdiv(g.Cout, 64);
+        std::vector<int> cands;
+        for (int sk : {1, 2, 4, 8, 16}) {
+          if (sk > 1 && (nsteps + sk - 1) / sk < 2) break;
+          if ((int64_t)base64 * sk > 16384) break;
+          cands.push_back(sk);
+        }
+        if (cands.empty()) {
+          ch = {0, 1};
+        } else if (stream_capturing(s) || getenv("RTHD_NO_AUTOTUNE")) {
+          // heuristic: smallest sk that brings the launch to >= 768
+          // blocks (or the largest allowed)
+          ch = {1, cands.back()};
+          for (int sk : cands) {
+            if (base64 * sk >= 768) { ch = {1, sk}; break; }
+          }
+        } else {
+          float best = 1e30f;
+          ch = {0, 1};
+          // big-kernel candidate measured via a recursive call with the
+          // cache primed (set, measure, restore)
+          for (int sk : cands) {
+            float t = time_usec([&] {
+              (void)conv_fwd_small(xc, wpk, scale, shift, skip, KH, KW,
+                                   stride, pad, Cout, act, sk);
+            }, s.stream());
+            if (t < best) { best = t; ch = {1, sk}; }
+          }
+          {  // time the big kernel: temporarily pin choice to big
+            std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+            g_conv_tune[key] = ConvChoice{0, 1};
+          }
+          float tbig = time_usec([&] {
+            (void)conv_fwd(xc, wpk, scale, shift, skip, KH, KW, stride,
+                           pad, Cout, act);
+          }, s.stream());
+          if (tbig < best) { best = tbig; ch = {0, 1}; }
+        }
+        std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+        g_conv_tune[key] = ch;
+      }
+      if (ch.small)
+        return conv_fwd_small(xc, wpk, scale, shift, skip, KH, KW, stride,
+                              pad, Cout, act, ch.splitk);
+    }
+
     auto zpage = torch::zeros({8}, xc.options());
     auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
     auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());

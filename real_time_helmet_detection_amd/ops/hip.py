@@ -211,7 +211,15 @@ class _ConvBNActFn(torch.autograd.Function):
             if has_skip:
                 dskip = dpre
         if has_bias and ctx.needs_input_grad[2]:
-            dbias = C.col_sum(dpre)
+            if use_bn and training:
+                # train-mode BN re-centers the conv output, so a constant
+                # bias shift cancels exactly: dL/dbias = sum(dpre) = 0
+                # analytically (sum of BN-backward input grads over the
+                # normalization axes is 0). Skip the reduction.
+                dbias = torch.zeros(dpre.shape[1], device=dpre.device,
+                                    dtype=torch.float32)
+            else:
+                dbias = C.col_sum(dpre)
 
         # wgrad — on the side stream, overlapping the dgrad launched above
         dw = None

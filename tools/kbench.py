@@ -52,7 +52,9 @@ def bench_conv(iters):
     shapes = [
         ('3x3 128ch @128^2 B16', 16, 128, 128, 128, 3, 1, 1),
         ('3x3 128ch @256^2 B16', 16, 128, 128, 256, 3, 1, 1),
+        ('3x3 128ch @64^2 B16', 16, 128, 128, 64, 3, 1, 1),
         ('3x3 128ch @32^2 B16', 16, 128, 128, 32, 3, 1, 1),
+        ('3x3 128ch @16^2 B16', 16, 128, 128, 16, 3, 1, 1),
         ('3x3 128ch @8^2 B16', 16, 128, 128, 8, 3, 1, 1),
         ('1x1 128ch @128^2 B16', 16, 128, 128, 128, 1, 1, 0),
     ]
@@ -63,10 +65,24 @@ def bench_conv(iters):
         wpk = C.pack_weights(w, False, True)
         ones = torch.ones(cout, device='cuda')
         zeros = torch.zeros(cout, device='cuda')
+        fl = 2 * B * hw * hw * cin * cout * k * k
+        # autotuned dispatch (first call measures variants and caches)
         ms = timeit(lambda: C.conv_fwd(x, wpk, ones, zeros, None, k, k,
                                        stride, pad, cout, 1), iters)
-        fl = 2 * B * hw * hw * cin * cout * k * k
-        print(f'conv  {name:26s} {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
+        print(f'conv(auto) {name:26s} {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
+        # explicit variants for the table
+        for sk in (1, 2, 4, 8, 16):
+            if sk > 9 * max(1, cin // 32):
+                continue
+            try:
+                ms = timeit(lambda: C.conv_fwd_small(
+                    x, wpk, ones, zeros, None, k, k, stride, pad, cout, 1,
+                    sk), iters)
+                print(f'conv(64,sk={sk:2d}) {name:22s} {ms*1e3:8.1f}us '
+                      f' {fl/ms/1e9:7.1f} TF')
+            except RuntimeError as e:
+                print(f'conv(64,sk={sk}) {name}: {e}')
+                break
 
 
 def bench_stem(iters):

@@ -61,11 +61,6 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
                              int64_t KH, int64_t KW, int64_t stride,
                              int64_t pad, int64_t Cout, int64_t act,
                              int64_t splitk);
-torch::Tensor stem_fwd(torch::Tensor x, torch::Tensor w,
-                       torch::Tensor scale, torch::Tensor shift,
-                       int64_t stride, int64_t pad, int64_t act);
-torch::Tensor stem_wgrad(torch::Tensor x, torch::Tensor dy, int64_t stride,
-                         int64_t pad);
 torch::Tensor stem_im2col(torch::Tensor x, int64_t KS, int64_t stride,
                           int64_t pad);
 torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
@@ -109,8 +104,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd_fp8", &rthd::conv_fwd_fp8);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("conv_fwd_small", &rthd::conv_fwd_small);
-  m.def("stem_fwd", &rthd::stem_fwd);
-  m.def("stem_wgrad", &rthd::stem_wgrad);
   m.def("stem_im2col", &rthd::stem_im2col);
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
@@ -158,8 +151,6 @@ TORCH_LIBRARY(rthd, m) {
   m.def("conv_fwd_fp8(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
         "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
         "int act) -> Tensor");
-  m.def("stem_fwd(Tensor x, Tensor w, Tensor scale, Tensor shift, "
-        "int stride, int pad, int act) -> Tensor");
   m.def("add_act_fwd(Tensor a, Tensor b, int act) -> Tensor");
   m.def("maxpool2x2(Tensor x) -> Tensor");
   m.def("avgpool2x2(Tensor x) -> Tensor");
@@ -175,7 +166,6 @@ TORCH_LIBRARY(rthd, m) {
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("conv_fwd", rthd::conv_fwd);
   m.impl("conv_fwd_fp8", rthd::conv_fwd_fp8);
-  m.impl("stem_fwd", rthd::stem_fwd);
   m.impl("add_act_fwd", rthd::add_act_fwd);
   m.impl("maxpool2x2", rthd::maxpool2x2_op);
   m.impl("avgpool2x2", rthd::avgpool2x2_op);

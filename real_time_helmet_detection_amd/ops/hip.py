@@ -120,17 +120,16 @@ class _ConvBNActFn(torch.autograd.Function):
         cout = weight.shape[0]
         dev = x.device
 
-        # bf16 stem: unfold to [px][152] once and run the MFMA conv as its
-        # 1x1 case — both faster than the direct VALU stem kernel AND the
+        # stem (any dtype): unfold to [px][152] once and run the MFMA conv
+        # as its 1x1 case — faster than a direct VALU stem kernel AND the
         # unfolded tensor is reused by the backward wgrad (which otherwise
-        # re-unfolds). fp32 stem keeps the direct kernel.
-        stem_col = is_stem and bf16 and kh == 7
-        wpk = None
+        # re-unfolds).
+        stem_col = is_stem and kh == 7
         if stem_col:
             xcol = C.stem_im2col(xc, kh, stride, pad)
-            wpk = C.pack_weights(_stem_col_weight(weight), False, True)
+            wpk = C.pack_weights(_stem_col_weight(weight), False, bf16)
             xc = xcol
-        elif not is_stem:
+        else:
             wpk = C.pack_weights(weight, False, bf16)
 
         ones, zeros = _ones_zeros(cout, dev)
@@ -142,10 +141,6 @@ class _ConvBNActFn(torch.autograd.Function):
                 assert sk is None
                 return ops.conv_fwd(xc, wpk, scale, shift, None, 1, 1, 1,
                                     0, cout, act)
-            if is_stem:
-                assert sk is None
-                return ops.stem_fwd(xc, weight, scale, shift, stride, pad,
-                                    act)
             return ops.conv_fwd(xc, wpk, scale, shift, sk, kh, kw, stride,
                                 pad, cout, act)
 
@@ -233,12 +228,13 @@ class _ConvBNActFn(torch.autograd.Function):
                     # forward: the wgrad is its 1x1 MFMA case; column
                     # t*3+ci -> dW[co][ci][t]
                     cout = dpre.shape[1]
-                    dwc = C.wgrad_bf16_fast(xc, dpre, 1, 1, 1, 0)
+                    if xc.dtype == torch.bfloat16:
+                        dwc = C.wgrad_bf16_fast(xc, dpre, 1, 1, 1, 0)
+                    else:
+                        dwc = C.wgrad(xc, dpre.float(), 1, 1, 1, 0)
                     dw = (dwc[:, :kh * kw * 3, 0, 0]
                           .reshape(cout, kh * kw, 3).permute(0, 2, 1)
                           .reshape(cout, 3, kh, kw).contiguous())
-                elif is_stem:
-                    dw = C.stem_wgrad(xc, dpre, stride, pad)
                 elif xc.dtype == torch.bfloat16:
                     # pad stray non-x8 channel counts (head Cout=6) so the
                     # aligned fast kernel runs; slice the result back
@@ -370,8 +366,7 @@ def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
             scale = torch.ones(cout, device=x.device, dtype=torch.float32)
             shift = bias_f
         if is_stem:
-            wpk = (_C().pack_weights(_stem_col_weight(weight), False, True)
-                   if bf16 else None)
+            wpk = _C().pack_weights(_stem_col_weight(weight), False, bf16)
         else:
             wpk = _C().pack_weights(weight, False, bf16)
         conv._rthd_infer_cache = (key, wpk, scale, shift)
@@ -380,15 +375,12 @@ def _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride, pad, is_stem):
 
     if is_stem:
         assert skc is None
-        if bf16:
-            # unfold + 1x1 MFMA conv (see _ConvBNActFn): faster than the
-            # direct VALU stem kernel. MUST go through the dispatcher op —
-            # a pybind call here would bake the warmup input's unfolded
-            # tensor into the traced export as a constant.
-            xcol = ops.stem_im2col(xc, kh, stride, pad)
-            return ops.conv_fwd(xcol, wpk, scale, shift, None, 1, 1, 1, 0,
-                                cout, act_code)
-        return ops.stem_fwd(xc, weight, scale, shift, stride, pad, act_code)
+        # unfold + 1x1 MFMA conv (see _ConvBNActFn). MUST go through the
+        # dispatcher op — a pybind call here would bake the warmup input's
+        # unfolded tensor into the traced export as a constant.
+        xcol = ops.stem_im2col(xc, kh, stride, pad)
+        return ops.conv_fwd(xcol, wpk, scale, shift, None, 1, 1, 1, 0,
+                            cout, act_code)
     return ops.conv_fwd(xc, wpk, scale, shift, skc, kh, kw, stride, pad,
                         cout, act_code)
 

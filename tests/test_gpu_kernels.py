@@ -416,15 +416,28 @@ def test_wgrad(cfg):
     assert rel_err(got, w.grad) < 1e-4
 
 
-def test_stem_fwd():
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_stem_im2col_conv(dtype):
+    """Unified stem path (any dtype): unfold + 1x1 MFMA conv == F.conv2d."""
     torch.manual_seed(13)
     x = torch.randn(2, 3, 64, 64)
     w = torch.randn(64, 3, 7, 7) * 0.1
     bias = torch.randn(64)
+    if dtype == torch.bfloat16:
+        x = x.to(dtype).float()
+        w = w.to(dtype).float()
     want = F.conv2d(x, w, bias, stride=2, padding=3)
+
+    from real_time_helmet_detection_amd.ops.hip import _stem_col_weight
+    xcol = _C().stem_im2col(to_gpu(x, dtype), 7, 2, 3)
+    assert xcol.shape[1] == 152
+    wpk = _C().pack_weights(_stem_col_weight(w.cuda()), False,
+                            dtype == torch.bfloat16)
     ones = torch.ones(64, device='cuda')
-    got = _C().stem_fwd(to_gpu(x), w.cuda(), ones, bias.cuda(), 2, 3, 0)
-    assert rel_err(got, want) < 1e-5
+    got = _C().conv_fwd(xcol, wpk, ones, bias.cuda(), None, 1, 1, 1, 0,
+                        64, 0)
+    tol = 0.02 if dtype == torch.bfloat16 else 1e-5
+    assert rel_err(got, want) < tol
 
 
 # -------------------------------------------------------------------- bn ---
@@ -518,7 +531,8 @@ def test_wgrad_bf16_fast(cfg):
 
 
 @pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
-def test_stem_wgrad(dtype):
+def test_stem_wgrad_im2col(dtype):
+    """Stem wgrad through the unfolded tensor (both engines)."""
     torch.manual_seed(19)
     x = torch.randn(2, 3, 64, 64)
     if dtype == torch.bfloat16:
@@ -529,7 +543,14 @@ def test_stem_wgrad(dtype):
     if dtype == torch.bfloat16:
         dy = dy.to(dtype).float()
     y.backward(dy)
-    got = _C().stem_wgrad(to_gpu(x, dtype), to_gpu(dy, dtype), 2, 3)
+
+    xcol = _C().stem_im2col(to_gpu(x, dtype), 7, 2, 3)
+    if dtype == torch.bfloat16:
+        dwc = _C().wgrad_bf16_fast(xcol, to_gpu(dy, dtype), 1, 1, 1, 0)
+    else:
+        dwc = _C().wgrad(xcol, to_gpu(dy, dtype), 1, 1, 1, 0)
+    got = (dwc[:, :49 * 3, 0, 0].reshape(64, 49, 3).permute(0, 2, 1)
+           .reshape(64, 3, 7, 7).contiguous())
     tol = 0.02 if dtype == torch.bfloat16 else 1e-4
     assert rel_err(got, w.grad) < tol
 

@@ -91,13 +91,19 @@ def bench_stem(iters):
     w = torch.randn(64, 3, 7, 7, device='cuda') * 0.05
     ones = torch.ones(64, device='cuda')
     zeros = torch.zeros(64, device='cuda')
-    ms = timeit(lambda: C.stem_fwd(x, w, ones, zeros, 2, 3, 1), iters)
     fl = 2 * 16 * 256 * 256 * 64 * 147
-    print(f'stem_fwd @512^2 B16           {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
+    wpk = C.pack_weights(
+        torch.nn.functional.pad(
+            w.permute(0, 2, 3, 1).reshape(64, -1), (0, 5)
+        ).reshape(64, -1, 1, 1), False, True)
+
+    def fwd_path():
+        xcol = C.stem_im2col(x, 7, 2, 3)
+        return C.conv_fwd(xcol, wpk, ones, zeros, None, 1, 1, 1, 0, 64, 1)
+    ms = timeit(fwd_path, iters)
+    print(f'stem_fwd(im2col) @512^2 B16   {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
     dy = torch.randn(16, 64, 256, 256, device='cuda',
                      dtype=torch.bfloat16).contiguous(memory_format=CL)
-    ms = timeit(lambda: C.stem_wgrad(x, dy, 2, 3), iters)
-    print(f'stem_wgrad(direct) @512^2 B16 {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
 
     def im2col_path():
         xcol = C.stem_im2col(x, 7, 2, 3)

@@ -577,8 +577,9 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
   const int64_t M = xc.numel() / C;
-  auto sum = torch::zeros({C}, xc.options().dtype(at::kFloat));
-  auto sumsq = torch::zeros({C}, xc.options().dtype(at::kFloat));
+  // fully overwritten by reduce_final_kernel (no zero-init kernels)
+  auto sum = torch::empty({C}, xc.options().dtype(at::kFloat));
+  auto sumsq = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto mean = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto rstd = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
@@ -665,8 +666,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor x,
   const int C = xc.size(1);
   const int64_t n = xc.numel();
   const int64_t M = n / C;
-  auto s1 = torch::zeros({C}, xc.options().dtype(at::kFloat));
-  auto s2 = torch::zeros({C}, xc.options().dtype(at::kFloat));
+  auto s1 = torch::empty({C}, xc.options().dtype(at::kFloat));
+  auto s2 = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto dx = torch::empty_like(xc);
   auto gm = gamma.to(at::kFloat).contiguous();
   auto bt = beta.to(at::kFloat).contiguous();
@@ -782,7 +783,7 @@ torch::Tensor col_sum(torch::Tensor x) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   const int C = xc.size(1);
   const int64_t M = xc.numel() / C;
-  auto sum = torch::zeros({C}, xc.options().dtype(at::kFloat));
+  auto sum = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
   run_colsum(xc, sum, nullptr, M, C, s);
   HIP_CHECK_LAST();

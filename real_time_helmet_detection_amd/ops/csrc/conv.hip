@@ -464,6 +464,22 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
                              int64_t pad, int64_t Cout, int64_t act,
                              int64_t splitk);
 
+// per-device cached 16-B zero page for the glds out-of-range source (a
+// fresh torch::zeros({8}) per conv call was ~85 FillFunctor launches per
+// training step). First use happens during eager warmup (never inside a
+// graph capture), so the allocation is from the regular allocator pool.
+const bf16* zero_page_bf16(const torch::Tensor& like) {
+  static std::mutex mu;
+  static std::unordered_map<int, torch::Tensor> pages;
+  const int dev = like.device().index();
+  std::lock_guard<std::mutex> lk(mu);
+  auto it = pages.find(dev);
+  if (it == pages.end())
+    it = pages.emplace(dev, torch::zeros({8},
+        like.options().dtype(at::kBFloat16))).first;
+  return reinterpret_cast<const bf16*>(it->second.data_ptr());
+}
+
 namespace {
 
 struct ConvChoice { int small; int splitk; };
@@ -561,8 +577,11 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                 "Cin=", g.Cin, " runs the f32 path");
 
     // -------- per-shape variant selection (autotune cache) --------
+    // tune when the 128-tile kernel can't fill 256 CUs x 2 waves, or when
+    // Cout <= 64 (the 128-wide N tile computes >= half dead columns, e.g.
+    // the 152->64 stem GEMM)
     const int big_blocks = (int)cdiv(g.M, 128) * (g.Coutp / 128);
-    if (big_blocks < 512) {  // big kernel can't fill 256 CUs x 2 waves
+    if (big_blocks < 512 || g.Cout <= 64) {
       ConvChoice ch{0, 1};
       bool have = false;
       const uint64_t key = conv_key(g);
@@ -621,11 +640,10 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                               pad, Cout, act, ch.splitk);
     }
 
-    auto zpage = torch::zeros({8}, xc.options());
     auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
     auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());
     auto* py = reinterpret_cast<bf16*>(y.data_ptr());
-    auto* pz = reinterpret_cast<const bf16*>(zpage.data_ptr());
+    const bf16* pz = zero_page_bf16(xc);
     const bf16* ps =
         has_skip ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
     if (has_skip)

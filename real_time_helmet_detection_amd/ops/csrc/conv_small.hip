@@ -23,6 +23,8 @@ namespace rthd {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
+const bf16* zero_page_bf16(const torch::Tensor& like);  // conv.hip
+
 DEV_INLINE int lds_off_bf16_s(int row, int k8) {
   return row * 64 + ((k8 ^ ((row >> 2) & 3)) << 4);
 }
@@ -281,7 +283,6 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
     sk_t = skip->to(at::kBFloat16).contiguous(
         at::MemoryFormat::ChannelsLast);
 
-  auto zpage = torch::zeros({8}, xc.options());
   const int nblkN = (int)cdiv(Cout, 64);
   const int Coutp64 = nblkN * 64;
   dim3 grid(cdiv(g.M, 64), nblkN, SK);
@@ -291,7 +292,7 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
   auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
   auto* pw = reinterpret_cast<const bf16*>(wpk.data_ptr());
   auto* py = reinterpret_cast<bf16*>(y.data_ptr());
-  auto* pz = reinterpret_cast<const bf16*>(zpage.data_ptr());
+  const bf16* pz = zero_page_bf16(xc);
   const bf16* ps =
       has_skip ? reinterpret_cast<const bf16*>(sk_t.data_ptr()) : nullptr;
 

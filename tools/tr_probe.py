@@ -54,7 +54,20 @@ def main():
                     '--genco', '/tmp/trp.hip', '-o', '/tmp/trp.hsaco'],
                    check=True)
     torch.cuda.init()
-    lib = ctypes.CDLL('/opt/rocm/lib/libamdhip64.so')
+    # use torch's bundled HIP runtime (the /opt/rocm copy can be a
+    # different version than the libhsa torch loaded -> dlopen fails)
+    libdir = os.path.join(os.path.dirname(torch.__file__), 'lib')
+    cand = [os.path.join(libdir, 'libamdhip64.so'), None,
+            '/opt/rocm/lib/libamdhip64.so']
+    lib = None
+    for c in cand:
+        try:
+            lib = ctypes.CDLL(c)
+            if hasattr(lib, 'hipModuleLoad'):
+                break
+        except OSError:
+            continue
+    assert lib is not None and hasattr(lib, 'hipModuleLoad')
     mod = ctypes.c_void_p()
     fn = ctypes.c_void_p()
     assert lib.hipModuleLoad(ctypes.byref(mod), b'/tmp/trp.hsaco') == 0

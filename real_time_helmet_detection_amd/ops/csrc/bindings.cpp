@@ -50,6 +50,12 @@ torch::Tensor conv_fwd_fp8(torch::Tensor x, torch::Tensor wpk,
                            c10::optional<torch::Tensor> skip,
                            int64_t KH, int64_t KW, int64_t stride,
                            int64_t pad, int64_t Cout, int64_t act);
+torch::Tensor conv_fwd_fp8r(torch::Tensor x, torch::Tensor wpk,
+                            torch::Tensor scale, torch::Tensor shift,
+                            c10::optional<torch::Tensor> skip,
+                            int64_t KH, int64_t KW, int64_t stride,
+                            int64_t pad, int64_t Cout, int64_t act,
+                            bool out_fp8);
 torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                        torch::Tensor scale, torch::Tensor shift,
                        c10::optional<torch::Tensor> skip,
@@ -102,6 +108,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_weights", &rthd::pack_weights);
   m.def("pack_weights_fp8", &rthd::pack_weights_fp8);
   m.def("conv_fwd_fp8", &rthd::conv_fwd_fp8);
+  m.def("conv_fwd_fp8r", &rthd::conv_fwd_fp8r);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("conv_fwd_small", &rthd::conv_fwd_small);
   m.def("stem_im2col", &rthd::stem_im2col);
@@ -151,6 +158,9 @@ TORCH_LIBRARY(rthd, m) {
   m.def("conv_fwd_fp8(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
         "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
         "int act) -> Tensor");
+  m.def("conv_fwd_fp8r(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
+        "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
+        "int act, bool out_fp8) -> Tensor");
   m.def("add_act_fwd(Tensor a, Tensor b, int act) -> Tensor");
   m.def("maxpool2x2(Tensor x) -> Tensor");
   m.def("avgpool2x2(Tensor x) -> Tensor");
@@ -166,6 +176,7 @@ TORCH_LIBRARY(rthd, m) {
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("conv_fwd", rthd::conv_fwd);
   m.impl("conv_fwd_fp8", rthd::conv_fwd_fp8);
+  m.impl("conv_fwd_fp8r", rthd::conv_fwd_fp8r);
   m.impl("add_act_fwd", rthd::add_act_fwd);
   m.impl("maxpool2x2", rthd::maxpool2x2_op);
   m.impl("avgpool2x2", rthd::avgpool2x2_op);

@@ -40,6 +40,18 @@ template <typename T> DEV_INLINE void stf(T* p, float v);
 template <> DEV_INLINE void stf<float>(float* p, float v) { *p = v; }
 template <> DEV_INLINE void stf<bf16>(bf16* p, float v) { *p = f2b(v); }
 
+// OCP e4m3 fp8 storage type (torch at::kFloat8_e4m3fn): math in f32 via
+// the gfx950 convert instructions; saturating on store.
+struct fp8e4 { unsigned char v; };
+
+template <> DEV_INLINE float ldf<fp8e4>(const fp8e4* p) {
+  return __builtin_amdgcn_cvt_f32_fp8((int)p->v, 0);
+}
+template <> DEV_INLINE void stf<fp8e4>(fp8e4* p, float v) {
+  p->v = (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(v, 0.f, 0, false)
+                         & 0xff);
+}
+
 // activation codes shared with python (ops/hip.py)
 enum ActKind : int { ACT_LINEAR = 0, ACT_RELU = 1, ACT_LRELU = 2 };
 

@@ -78,8 +78,22 @@ torch::Tensor add_act_fwd(torch::Tensor a, torch::Tensor b, int64_t act) {
       hipLaunchKernelGGL((add_act_fwd_kernel<bf16, 1>),
           dim3(ew_grid(n, block)), dim3(block), 0, stream,
           pa, pb, py, n, (int)act);
+  } else if (ac.scalar_type() == at::kFloat8_e4m3fn) {
+    constexpr int VEC = 16;  // fp8-resident inference chain
+    auto* pa = reinterpret_cast<const fp8e4*>(ac.data_ptr());
+    auto* pb = reinterpret_cast<const fp8e4*>(bc.data_ptr());
+    auto* py = reinterpret_cast<fp8e4*>(y.data_ptr());
+    if (n % VEC == 0)
+      hipLaunchKernelGGL((add_act_fwd_kernel<fp8e4, VEC>),
+          dim3(ew_grid(n / VEC, block)), dim3(block), 0, stream,
+          pa, pb, py, n, (int)act);
+    else
+      hipLaunchKernelGGL((add_act_fwd_kernel<fp8e4, 1>),
+          dim3(ew_grid(n, block)), dim3(block), 0, stream,
+          pa, pb, py, n, (int)act);
   } else {
-    TORCH_CHECK(ac.scalar_type() == at::kFloat, "add_act: bf16/f32 only");
+    TORCH_CHECK(ac.scalar_type() == at::kFloat,
+                "add_act: bf16/f32/e4m3 only");
     constexpr int VEC = 4;
     if (n % VEC == 0)
       hipLaunchKernelGGL((add_act_fwd_kernel<float, VEC>),

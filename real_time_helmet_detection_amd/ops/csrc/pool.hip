@@ -144,8 +144,12 @@ static std::tuple<int, int, int, int> nhwc_dims(const torch::Tensor& t) {
   if ((tensor).scalar_type() == at::kBFloat16) {                             \
     using scalar_t = bf16;                                                   \
     fn                                                                       \
+  } else if ((tensor).scalar_type() == at::kFloat8_e4m3fn) {                 \
+    using scalar_t = fp8e4;  /* fp8-resident inference chain */              \
+    fn                                                                       \
   } else {                                                                   \
-    TORCH_CHECK((tensor).scalar_type() == at::kFloat, "bf16/f32 only");      \
+    TORCH_CHECK((tensor).scalar_type() == at::kFloat,                        \
+                "bf16/f32/e4m3 only");                                       \
     using scalar_t = float;                                                  \
     fn                                                                       \
   }

@@ -75,29 +75,28 @@ torch::Tensor upsample2x_add_fwd(torch::Tensor x,
     TORCH_CHECK(sc.size(2) == 2 * H && sc.size(3) == 2 * W &&
                 sc.size(1) == C, "upsample2x_add: skip shape mismatch");
   }
-  if (xc.scalar_type() == at::kBFloat16) {
-    auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
-    auto* py = reinterpret_cast<bf16*>(y.data_ptr());
-    const bf16* ps = has_skip ? reinterpret_cast<const bf16*>(sc.data_ptr())
-                              : nullptr;
-    if (has_skip)
-      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<bf16, true>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C);
-    else
-      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<bf16, false>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C);
-  } else {
-    TORCH_CHECK(xc.scalar_type() == at::kFloat, "bf16/f32 only");
-    auto* px = xc.data_ptr<float>();
-    auto* py = y.data_ptr<float>();
-    const float* ps = has_skip ? sc.data_ptr<float>() : nullptr;
-    if (has_skip)
-      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<float, true>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C);
-    else
-      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<float, false>),
-          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C);
+#define RTHD_UPS_LAUNCH(T)                                                 \
+  {                                                                        \
+    auto* px = reinterpret_cast<const T*>(xc.data_ptr());                  \
+    auto* py = reinterpret_cast<T*>(y.data_ptr());                         \
+    const T* ps = has_skip                                                 \
+        ? reinterpret_cast<const T*>(sc.data_ptr()) : nullptr;             \
+    if (has_skip)                                                          \
+      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<T, true>),             \
+          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C); \
+    else                                                                   \
+      hipLaunchKernelGGL((upsample2x_add_fwd_kernel<T, false>),            \
+          dim3(ew_grid(n, 256)), dim3(256), 0, s, px, ps, py, B, H, W, C); \
   }
+  if (xc.scalar_type() == at::kBFloat16) {
+    RTHD_UPS_LAUNCH(bf16)
+  } else if (xc.scalar_type() == at::kFloat8_e4m3fn) {
+    RTHD_UPS_LAUNCH(fp8e4)
+  } else {
+    TORCH_CHECK(xc.scalar_type() == at::kFloat, "bf16/f32/e4m3 only");
+    RTHD_UPS_LAUNCH(float)
+  }
+#undef RTHD_UPS_LAUNCH
   HIP_CHECK_LAST();
   return y;
 }

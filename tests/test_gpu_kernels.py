@@ -383,6 +383,58 @@ def test_conv_autotune_dispatch_matches_big():
     assert rel_err(got2, want) < 0.03
 
 
+@pytest.mark.parametrize('out_fp8', [True, False])
+def test_conv_fwd_fp8_resident(out_fp8):
+    """fp8-resident conv (e4m3 in, e4m3/bf16 out, fused epilogue + fp8
+    skip) vs torch conv on the dequantized values."""
+    torch.manual_seed(21)
+    x8 = (torch.randn(4, 32, 16, 16) * 0.5).to(torch.float8_e4m3fn)
+    x = x8.float()
+    w = torch.randn(32, 32, 3, 3) * 0.1
+    sw = w.abs().amax(dim=(1, 2, 3)).clamp(min=1e-8) / 240.0
+    w8 = (w / sw.view(-1, 1, 1, 1)).to(torch.float8_e4m3fn).float() \
+        * sw.view(-1, 1, 1, 1)
+    sk8 = (torch.randn(4, 32, 16, 16) * 0.5).to(torch.float8_e4m3fn)
+    sc = torch.rand(32) + 0.5
+    sh = torch.randn(32) * 0.1
+    want = F.relu(F.conv2d(x, w8, None, padding=1) * sc.view(1, -1, 1, 1)
+                  + sh.view(1, -1, 1, 1) + sk8.float())
+
+    wpk = _C().pack_weights_fp8((w / sw.view(-1, 1, 1, 1)).cuda())
+    got = _C().conv_fwd_fp8r(
+        x8.cuda().contiguous(memory_format=CL), wpk,
+        (sc * sw).cuda(), sh.cuda(),
+        sk8.cuda().contiguous(memory_format=CL),
+        3, 3, 1, 1, 32, 1, out_fp8)
+    assert got.dtype == (torch.float8_e4m3fn if out_fp8
+                         else torch.bfloat16)
+    assert rel_err(got.float(), want) < (0.08 if out_fp8 else 0.05)
+
+
+def test_fp8_elementwise_ops():
+    """fp8 instantiations of pool/upsample/add (the fp8-resident chain's
+    glue ops) vs f32 on dequantized values."""
+    torch.manual_seed(22)
+    x8 = (torch.randn(2, 16, 8, 8) * 2).to(torch.float8_e4m3fn)
+    x = x8.float()
+    g8 = x8.cuda().contiguous(memory_format=CL)
+
+    got = _C().pool2x2_fwd(g8, True, False)[0]
+    assert got.dtype == torch.float8_e4m3fn
+    assert rel_err(got.float(), F.max_pool2d(x, 2, 2)) < 1e-6  # exact copy
+
+    sk8 = (torch.randn(2, 16, 16, 16)).to(torch.float8_e4m3fn)
+    got = _C().upsample2x_add_fwd(g8, sk8.cuda().contiguous(
+        memory_format=CL))
+    want = F.interpolate(x, scale_factor=2, mode='nearest') + sk8.float()
+    assert rel_err(got.float(), want) < 0.08
+
+    b8 = (torch.randn(2, 16, 8, 8)).to(torch.float8_e4m3fn)
+    got = _C().add_act_fwd(g8, b8.cuda().contiguous(memory_format=CL), 1)
+    want = F.relu(x + b8.float())
+    assert rel_err(got.float(), want) < 0.08
+
+
 def test_dgrad_via_swapped_pack():
     # conv3x3 s1 p1: dX = conv(dY, rot180(W) transposed)
     torch.manual_seed(11)

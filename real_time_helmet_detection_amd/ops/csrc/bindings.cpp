@@ -45,11 +45,6 @@ torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
 
 torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16);
 torch::Tensor pack_weights_fp8(torch::Tensor w);
-torch::Tensor conv_fwd_fp8(torch::Tensor x, torch::Tensor wpk,
-                           torch::Tensor scale, torch::Tensor shift,
-                           c10::optional<torch::Tensor> skip,
-                           int64_t KH, int64_t KW, int64_t stride,
-                           int64_t pad, int64_t Cout, int64_t act);
 torch::Tensor conv_fwd_fp8r(torch::Tensor x, torch::Tensor wpk,
                             torch::Tensor scale, torch::Tensor shift,
                             c10::optional<torch::Tensor> skip,
@@ -107,7 +102,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nms_fwd", &rthd::nms_fwd);
   m.def("pack_weights", &rthd::pack_weights);
   m.def("pack_weights_fp8", &rthd::pack_weights_fp8);
-  m.def("conv_fwd_fp8", &rthd::conv_fwd_fp8);
   m.def("conv_fwd_fp8r", &rthd::conv_fwd_fp8r);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("conv_fwd_small", &rthd::conv_fwd_small);
@@ -155,9 +149,6 @@ TORCH_LIBRARY(rthd, m) {
   m.def("conv_fwd(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
         "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
         "int act) -> Tensor");
-  m.def("conv_fwd_fp8(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
-        "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
-        "int act) -> Tensor");
   m.def("conv_fwd_fp8r(Tensor x, Tensor wpk, Tensor scale, Tensor shift, "
         "Tensor? skip, int kh, int kw, int stride, int pad, int cout, "
         "int act, bool out_fp8) -> Tensor");
@@ -175,7 +166,6 @@ TORCH_LIBRARY(rthd, m) {
 
 TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("conv_fwd", rthd::conv_fwd);
-  m.impl("conv_fwd_fp8", rthd::conv_fwd_fp8);
   m.impl("conv_fwd_fp8r", rthd::conv_fwd_fp8r);
   m.impl("add_act_fwd", rthd::add_act_fwd);
   m.impl("maxpool2x2", rthd::maxpool2x2_op);

@@ -675,216 +675,6 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
 
 
 
-// ------------------------------ fp8 inference -------------------------------
-// OCP e4m3 MFMA path (v_mfma_f32_16x16x32_fp8_fp8, 2x the bf16 rate).
-// Weights are pre-scaled per output channel python-side and packed to fp8;
-// activations arrive bf16 and are converted to fp8 IN STAGING (saturating),
-// so no fp8 tensors flow between layers. Inference only (BASELINE config 5).
-
-using fp8x8 = long long;  // 8 e4m3 bytes = one MFMA operand register pair
-
-// LDS tile: [128 rows][32 k] fp8, rows padded to 48 B (16-B write align +
-// conflict-free b64 fragment reads: bank = (12*row + 2*k8) % 64).
-DEV_INLINE int lds_off_fp8(int row, int chunk16) {
-  return row * 48 + (chunk16 << 4);
-}
-DEV_INLINE int lds_off_fp8_frag(int row, int k8) {
-  return row * 48 + (k8 << 3);
-}
-
-template <bool HIWORD>
-DEV_INLINE unsigned cvt2_fp8(float lo, float hi, unsigned old) {
-  return __builtin_amdgcn_cvt_pk_fp8_f32(lo, hi, old, HIWORD);
-}
-
-template <bool HAS_SKIP>
-__global__ __launch_bounds__(256)
-void conv_fwd_fp8_kernel(const bf16* __restrict__ x,
-                         const unsigned char* __restrict__ wpk,
-                         const float* __restrict__ scale,
-                         const float* __restrict__ shift,
-                         const bf16* __restrict__ skip,
-                         bf16* __restrict__ y,
-                         ConvGeo g, int act) {
-  const int mblk = blockIdx.x;
-  const int nblk = blockIdx.y;
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int wr = wid >> 1, wc = wid & 1;
-
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* ldsA = smem;                    // 6 KB
-  char* ldsB = smem + 128 * 48;         // 6 KB
-
-  f32x4 acc[4][4] = {};
-
-  // staging: one row (16 k-elems = one 16-B chunk of fp8) per thread:
-  // st_row = tid>>1 (128 rows), st_c16 = tid&1 (two 16-elem chunks)
-  const int st_row = tid >> 1;
-  const int st_c16 = tid & 1;
-
-  int am, ab, ay, ax;
-  {
-    const int m = mblk * 128 + st_row;
-    am = m;
-    const int mm = m < g.M ? m : 0;
-    ab = mm / (g.Ho * g.Wo);
-    const int r = mm % (g.Ho * g.Wo);
-    ay = r / g.Wo;
-    ax = r % g.Wo;
-  }
-
-  const int nsteps = g.KH * g.KW * (g.Cinp / 32);
-  const int kc_per_tap = g.Cinp / 32;
-
-  for (int step = 0; step < nsteps; ++step) {
-    const int t = step / kc_per_tap;
-    const int kb = step % kc_per_tap;
-    const int dy = t / g.KW - g.pad;
-    const int dx = t % g.KW - g.pad;
-
-    if (step) __syncthreads();  // previous fragment reads complete
-
-    // ---- stage A: load 16 bf16, convert to 16 fp8, one b128 write ----
-    {
-      const int iy = ay * g.stride + dy;
-      const int ix = ax * g.stride + dx;
-      const int c0 = kb * 32 + st_c16 * 16;
-      uint4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
-      const bool val = am < g.M && iy >= 0 && iy < g.H && ix >= 0 &&
-                       ix < g.W && c0 < g.Cin;
-      const int64_t off = val
-          ? (((int64_t)ab * g.H + iy) * g.W + ix) * g.Cin + c0 : 0;
-      v0 = *reinterpret_cast<const uint4*>(x + off);
-      v1 = *reinterpret_cast<const uint4*>(x + off + 8);
-      if (!val) { v0 = uint4{0, 0, 0, 0}; v1 = uint4{0, 0, 0, 0}; }
-      const bf16* b0 = reinterpret_cast<const bf16*>(&v0);
-      const bf16* b1 = reinterpret_cast<const bf16*>(&v1);
-      unsigned o[4] = {};
-#pragma unroll
-      for (int q = 0; q < 2; ++q) {
-        o[q] = cvt2_fp8<false>(b2f(b0[4 * q + 0]), b2f(b0[4 * q + 1]), o[q]);
-        o[q] = cvt2_fp8<true>(b2f(b0[4 * q + 2]), b2f(b0[4 * q + 3]), o[q]);
-        o[2 + q] = cvt2_fp8<false>(b2f(b1[4 * q + 0]), b2f(b1[4 * q + 1]),
-                                   o[2 + q]);
-        o[2 + q] = cvt2_fp8<true>(b2f(b1[4 * q + 2]), b2f(b1[4 * q + 3]),
-                                  o[2 + q]);
-      }
-      *reinterpret_cast<uint4*>(ldsA + lds_off_fp8(st_row, st_c16)) =
-          *reinterpret_cast<uint4*>(o);
-    }
-    // ---- stage B: weights already fp8 ----
-    {
-      const int64_t src_off =
-          ((int64_t)t * g.Coutp + nblk * 128 + st_row) * g.Cinp + kb * 32 +
-          st_c16 * 16;
-      *reinterpret_cast<uint4*>(ldsB + lds_off_fp8(st_row, st_c16)) =
-          *reinterpret_cast<const uint4*>(wpk + src_off);
-    }
-    __syncthreads();
-
-    const int arow_base = wr * 64 + (lane & 15);
-    const int brow_base = wc * 64 + (lane & 15);
-    const int k8 = lane >> 4;
-    fp8x8 afrag[4], bfrag[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      afrag[i] = *reinterpret_cast<const fp8x8*>(
-          ldsA + lds_off_fp8_frag(arow_base + 16 * i, k8));
-      bfrag[i] = *reinterpret_cast<const fp8x8*>(
-          ldsB + lds_off_fp8_frag(brow_base + 16 * i, k8));
-    }
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
-  }
-
-  // preload the 4 per-lane scale/shift pairs ONCE (the per-store scalar
-  // loads serialized the epilogue: 65 dependent vmcnt(0) waits in the .s)
-  const int col0 = nblk * 128 + wc * 64 + (lane & 15);
-  float esc[4], esh[4];
-#pragma unroll
-  for (int ni = 0; ni < 4; ++ni) {
-    const int c = col0 + ni * 16;
-    esc[ni] = c < g.Cout ? scale[c] : 0.f;
-    esh[ni] = c < g.Cout ? shift[c] : 0.f;
-  }
-  const int row_in_frag = (lane >> 4) * 4;
-#pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = mblk * 128 + wr * 64 + mi * 16 + row_in_frag + r;
-      if (m >= g.M) continue;
-#pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const int c = col0 + ni * 16;
-        if (c >= g.Cout) continue;
-        float v = acc[mi][ni][r];
-        v = v * esc[ni] + esh[ni];
-        if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
-        v = apply_act(v, act);
-        stf(&y[(int64_t)m * g.Cout + c], v);
-      }
-    }
-  }
-}
-
-torch::Tensor conv_fwd_fp8(torch::Tensor x, torch::Tensor wpk,
-                           torch::Tensor scale, torch::Tensor shift,
-                           c10::optional<torch::Tensor> skip,
-                           int64_t KH, int64_t KW, int64_t stride,
-                           int64_t pad, int64_t Cout, int64_t act) {
-  auto xc = x.to(at::kBFloat16).contiguous(at::MemoryFormat::ChannelsLast);
-  ConvGeo g;
-  g.B = xc.size(0);
-  g.Cin = xc.size(1);
-  g.H = xc.size(2);
-  g.W = xc.size(3);
-  g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
-  g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
-  g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
-  g.Cout = Cout;
-  g.Cinp = (int)cdiv(g.Cin, 32) * 32;
-  g.Coutp = (int)cdiv(Cout, 128) * 128;
-  g.M = g.B * g.Ho * g.Wo;
-  TORCH_CHECK(g.Cin % 16 == 0, "conv_fwd_fp8: Cin % 16 required");
-  TORCH_CHECK(wpk.scalar_type() == at::kFloat8_e4m3fn &&
-              wpk.size(1) == g.Coutp && wpk.size(2) == g.Cinp);
-
-  auto y = torch::empty({g.B, (int64_t)g.Cout, g.Ho, g.Wo},
-                        xc.options().memory_format(
-                            at::MemoryFormat::ChannelsLast));
-  auto sc = scale.to(at::kFloat).contiguous();
-  auto sh = shift.to(at::kFloat).contiguous();
-  const bool has_skip = skip.has_value();
-  torch::Tensor sk;
-  if (has_skip)
-    sk = skip->to(at::kBFloat16).contiguous(at::MemoryFormat::ChannelsLast);
-
-  dim3 grid(cdiv(g.M, 128), g.Coutp / 128);
-  auto s = at::cuda::getCurrentCUDAStream();
-  const size_t lds = 2 * 128 * 48;
-  auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
-  auto* pw = reinterpret_cast<const unsigned char*>(wpk.data_ptr());
-  auto* py = reinterpret_cast<bf16*>(y.data_ptr());
-  const bf16* ps = has_skip
-      ? reinterpret_cast<const bf16*>(sk.data_ptr()) : nullptr;
-  if (has_skip)
-    hipLaunchKernelGGL((conv_fwd_fp8_kernel<true>), grid, dim3(256), lds, s,
-        px, pw, sc.data_ptr<float>(), sh.data_ptr<float>(), ps, py, g,
-        (int)act);
-  else
-    hipLaunchKernelGGL((conv_fwd_fp8_kernel<false>), grid, dim3(256), lds, s,
-        px, pw, sc.data_ptr<float>(), sh.data_ptr<float>(), ps, py, g,
-        (int)act);
-  HIP_CHECK_LAST();
-  return y;
-}
 
 // pack pre-scaled fp32 weights to fp8 e4m3 [T][Coutp][Cinp]
 __global__ void pack_weights_fp8_kernel(const float* __restrict__ w,
@@ -913,7 +703,8 @@ torch::Tensor pack_weights_fp8(torch::Tensor w) {
   const int KH = wc.size(2), KW = wc.size(3);
   const int T_ = KH * KW;
   const int Rp = (int)cdiv(Cout, 128) * 128;
-  const int Kp = (int)cdiv(Cin, 32) * 32;
+  // K pads to 128: one mfma_scale_f32_16x16x128 K-block per tap chunk
+  const int Kp = (int)cdiv(Cin, 128) * 128;
   auto pk = torch::empty({T_, Rp, Kp},
                          wc.options().dtype(at::kFloat8_e4m3fn));
   const int64_t n = (int64_t)T_ * Rp * Kp;

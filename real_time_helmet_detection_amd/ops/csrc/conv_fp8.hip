@@ -1,17 +1,22 @@
-// fp8-resident implicit-GEMM convolution (OCP e4m3) for gfx950.
+// fp8-resident implicit-GEMM convolution (OCP e4m3) for gfx950, built on
+// the K=128 block-scaled MFMA.
 //
-// BASELINE config 5. Round 1's fp8 kernel converted bf16 activations to
-// fp8 inside a single-buffered staging loop (2 barriers/step, no
-// prefetch) and measured SLOWER than the pipelined bf16 kernel despite
-// the 2x MFMA rate. This version keeps activations fp8 BETWEEN layers
-// (the epilogue emits e4m3; pools/upsamples/adds have fp8
-// instantiations), so staging is the same async global_load_lds 3-deep
-// ring as conv.hip with zero conversion work and HALF the bytes:
-// A tile = 128 px x 32 ch fp8 = 4 KB (32-B rows, lane-linear glds image,
-// b64 fragment reads accept the 2-way row-alias conflict), B tile from
-// pack_weights_fp8. MFMA: v_mfma_f32_16x16x32_fp8_fp8, fp32 accumulate,
-// fused scale/shift/act(+fp8 skip) epilogue, output e4m3 (mid-network)
-// or bf16 (heads feeding the decode).
+// gfx950's non-scaled fp8 MFMA (16x16x32) runs at the BF16 rate — only
+// the MX-scaled forms reach the 2x fp8 rate (guide §"MFMA shapes":
+// mfma_scale_f32_16x16x128_f8f6f4, >=4.6 PF dense with fmt=e4m3). This
+// kernel uses it with unit E8M0 scales (0x7F = 2^0), which reduces it to
+// a plain fp8 GEMM at the scaled-instruction rate. The probe
+// (tools/mfma_scale_probe.py) verified that any CONSISTENT per-lane k
+// ordering of A and B is valid under unit scales (the lane-element
+// products pair positionally), so A and B tiles share one staging map.
+//
+// Activations stay e4m3 BETWEEN layers (epilogue emits e4m3; pools /
+// upsamples / adds have fp8 instantiations; heads drop to bf16 for the
+// decode). Staging is async global_load_lds, double-buffered: one K-step
+// = ONE tap x 128 channels (A tile 128 px x 128 fp8 = 16 KB; B tile
+// 128 couts x 128 = 16 KB; 64 KB LDS -> 2 blocks/CU), per-thread 4+4
+// glds per step, source chunk pre-swizzled (chunk ^ (row&7)) so the
+// lane-linear glds image matches the conflict-reduced fragment reads.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
@@ -19,20 +24,24 @@
 namespace rthd {
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
-using fp8x8 = long long;
+using i32x8 = __attribute__((ext_vector_type(8))) int;
 
 const bf16* zero_page_bf16(const torch::Tensor& like);  // conv.hip
 
 struct ConvGeoF8 {
   int B, H, W, Cin, Ho, Wo, Cout;
   int KH, KW, stride, pad;
-  int Cinp;   // padded to 32
+  int Cinp;   // padded to 128
   int Coutp;  // pack_weights_fp8 pads rows to 128
   int M;
 };
 
 typedef __attribute__((address_space(3))) void lds_void_f8;
 typedef __attribute__((address_space(1))) const void glb_void_f8;
+
+DEV_INLINE int f8_off(int row, int chunk) {          // 128-B rows, 16-B chunks
+  return row * 128 + ((chunk ^ (row & 7)) << 4);
+}
 
 template <bool HAS_SKIP, typename OUT_T>
 __global__ __launch_bounds__(256)
@@ -52,60 +61,76 @@ void conv_fwd_fp8r_kernel(const fp8e4* __restrict__ x,
   const int wr = wid >> 1, wc = wid & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* lds = smem;  // 3 x (A 4KB | B 4KB)
+  char* lds = smem;  // 2 x (A 16KB | B 16KB)
 
   f32x4 acc[4][4] = {};
 
-  // staging: 128 rows x 2 chunks(16 B) per tile, one chunk per thread;
-  // glds is lane-linear (wave w covers bytes w*1024..+1023 = rows
-  // w*32 + lane/2, chunk lane&1) — matches this row/chunk assignment
-  const int st_row = tid >> 1;
-  const int st_c16 = tid & 1;
-  const int wbase = wid * 1024;
+  // staging: per tile, glds j of wave w covers rows (j*4 + w)*8 + lane/8,
+  // chunk lane&7 (lane-linear 1024-B wave writes); the SOURCE channel
+  // chunk is pre-swizzled so the image equals f8_off's layout
+  const int st_chunk = lane & 7;
 
-  int am[1], ab, ay, ax;
-  {
-    const int m = mblk * 128 + st_row;
-    am[0] = m;
+  // per-j pixel decomposition for the A rows this thread stages
+  int am[4], ab[4], ay[4], ax[4], asw[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int row = (j * 4 + wid) * 8 + (lane >> 3);
+    const int m = mblk * 128 + row;
+    am[j] = m;
+    asw[j] = (st_chunk ^ (row & 7)) * 16;  // source channel offset
     const int mm = m < g.M ? m : 0;
-    ab = mm / (g.Ho * g.Wo);
+    ab[j] = mm / (g.Ho * g.Wo);
     const int r = mm % (g.Ho * g.Wo);
-    ay = r / g.Wo;
-    ax = r % g.Wo;
+    ay[j] = r / g.Wo;
+    ax[j] = r % g.Wo;
   }
 
-  const int kc = g.Cinp / 32;
+  const int kc = g.Cinp / 128;
   const int taps = g.KH * g.KW;
   const int nsteps = taps * kc;
 
   int is_step = 0;
   int is_t = 0, is_kb = 0;
-  const fp8e4* aptr;
-  const fp8e4* bptr;
-  bool avalid;
+  const fp8e4* aptr[4];
+  const fp8e4* bptr[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int row = (j * 4 + wid) * 8 + (lane >> 3);
+    bptr[j] = wpk + ((int64_t)nblk * 128 + row) * g.Cinp + asw[j];
+  }
+  const int64_t btap = (int64_t)g.Coutp * g.Cinp;
+  bool avalid[4];
   auto tap_setup = [&]() {
     const int dy_ = is_t / g.KW - g.pad;
     const int dx_ = is_t % g.KW - g.pad;
-    const int iy = ay * g.stride + dy_;
-    const int ix = ax * g.stride + dx_;
-    avalid = am[0] < g.M && iy >= 0 && iy < g.H && ix >= 0 && ix < g.W;
-    aptr = avalid
-        ? x + (((int64_t)ab * g.H + iy) * g.W + ix) * g.Cin + st_c16 * 16
-        : reinterpret_cast<const fp8e4*>(zpage);
-    bptr = wpk + ((int64_t)is_t * g.Coutp + nblk * 128 + st_row) * g.Cinp +
-        st_c16 * 16;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int iy = ay[j] * g.stride + dy_;
+      const int ix = ax[j] * g.stride + dx_;
+      avalid[j] = am[j] < g.M && iy >= 0 && iy < g.H && ix >= 0 &&
+                  ix < g.W;
+      aptr[j] = avalid[j]
+          ? x + (((int64_t)ab[j] * g.H + iy) * g.W + ix) * g.Cin + asw[j]
+          : zpage;
+    }
   };
   tap_setup();
 
   auto issue_step = [&]() {
-    char* base = lds + (is_step % 3) * 8192;
-    const int cb = is_kb * 32;
-    const fp8e4* a0 = (avalid && cb + st_c16 * 16 < g.Cin)
-        ? aptr + cb : reinterpret_cast<const fp8e4*>(zpage);
-    __builtin_amdgcn_global_load_lds((glb_void_f8*)a0,
-        (lds_void_f8*)(base + wbase), 16, 0, 0);
-    __builtin_amdgcn_global_load_lds((glb_void_f8*)(bptr + cb),
-        (lds_void_f8*)(base + 4096 + wbase), 16, 0, 0);
+    char* base = lds + (is_step & 1) * 32768;
+    const int cb = is_kb * 128;
+    const int64_t boff = (int64_t)is_t * btap + cb;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const fp8e4* a0 = (avalid[j] && cb + asw[j] < g.Cin)
+          ? aptr[j] + cb : zpage;
+      __builtin_amdgcn_global_load_lds((glb_void_f8*)a0,
+          (lds_void_f8*)(base + (j * 4 + wid) * 1024 + (lane & 63) * 16),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds((glb_void_f8*)(bptr[j] + boff),
+          (lds_void_f8*)(base + 16384 + (j * 4 + wid) * 1024 +
+                         (lane & 63) * 16), 16, 0, 0);
+    }
     ++is_step;
     if (++is_kb == kc) {
       is_kb = 0;
@@ -114,43 +139,43 @@ void conv_fwd_fp8r_kernel(const fp8e4* __restrict__ x,
   };
 
   issue_step();
-  if (nsteps > 1) {
-    issue_step();
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2) : "memory");
-  } else {
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(0) : "memory");
-  }
+  asm volatile("s_waitcnt vmcnt(0)" ::"i"(0) : "memory");
   __builtin_amdgcn_s_barrier();
 
   for (int step = 0; step < nsteps; ++step) {
-    char* A = lds + (step % 3) * 8192;
-    char* B = A + 4096;
-    if (step + 2 < nsteps) issue_step();
+    char* A = lds + (step & 1) * 32768;
+    char* B = A + 16384;
+    if (step + 1 < nsteps) issue_step();
 
     const int arow_base = wr * 64 + (lane & 15);
     const int brow_base = wc * 64 + (lane & 15);
-    const int k8 = lane >> 4;
-    fp8x8 afrag[4], bfrag[4];
+    const int c2 = (lane >> 4) * 2;  // two 16-B chunks = 32 k per lane
+    i32x8 afrag[4], bfrag[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      afrag[i] = *reinterpret_cast<const fp8x8*>(
-          A + (arow_base + 16 * i) * 32 + k8 * 8);
-      bfrag[i] = *reinterpret_cast<const fp8x8*>(
-          B + (brow_base + 16 * i) * 32 + k8 * 8);
+      uint4 lo = *reinterpret_cast<const uint4*>(
+          A + f8_off(arow_base + 16 * i, c2));
+      uint4 hi = *reinterpret_cast<const uint4*>(
+          A + f8_off(arow_base + 16 * i, c2 + 1));
+      afrag[i] = i32x8{(int)lo.x, (int)lo.y, (int)lo.z, (int)lo.w,
+                       (int)hi.x, (int)hi.y, (int)hi.z, (int)hi.w};
+      lo = *reinterpret_cast<const uint4*>(
+          B + f8_off(brow_base + 16 * i, c2));
+      hi = *reinterpret_cast<const uint4*>(
+          B + f8_off(brow_base + 16 * i, c2 + 1));
+      bfrag[i] = i32x8{(int)lo.x, (int)lo.y, (int)lo.z, (int)lo.w,
+                       (int)hi.x, (int)hi.y, (int)hi.z, (int)hi.w};
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0,
+            0, 0x7F7F7F7F, 0, 0x7F7F7F7F);
 
-    if (step + 2 < nsteps + 1) {
-      if (step + 2 < nsteps)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2) : "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(0) : "memory");
-    }
+    if (step + 1 < nsteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::"i"(0) : "memory");
     __builtin_amdgcn_s_barrier();
   }
 
@@ -183,8 +208,8 @@ void conv_fwd_fp8r_kernel(const fp8e4* __restrict__ x,
   }
 }
 
-// x: e4m3 channels_last; wpk from pack_weights_fp8 (per-cout pre-scaled);
-// out_fp8 selects e4m3 (mid-network) vs bf16 (heads) output.
+// x: e4m3 channels_last; wpk from pack_weights_fp8 (per-cout pre-scaled,
+// K padded to 128); out_fp8 selects e4m3 (mid-network) vs bf16 (heads).
 torch::Tensor conv_fwd_fp8r(torch::Tensor x, torch::Tensor wpk,
                             torch::Tensor scale, torch::Tensor shift,
                             c10::optional<torch::Tensor> skip,
@@ -203,7 +228,7 @@ torch::Tensor conv_fwd_fp8r(torch::Tensor x, torch::Tensor wpk,
   g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
   g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
   g.Cout = Cout;
-  g.Cinp = (int)cdiv(g.Cin, 32) * 32;
+  g.Cinp = (int)cdiv(g.Cin, 128) * 128;
   g.Coutp = (int)cdiv(Cout, 128) * 128;
   g.M = g.B * g.Ho * g.Wo;
   TORCH_CHECK(g.Cin % 16 == 0, "conv_fwd_fp8r: Cin % 16 required");
@@ -227,7 +252,7 @@ torch::Tensor conv_fwd_fp8r(torch::Tensor x, torch::Tensor wpk,
 
   dim3 grid(cdiv(g.M, 128), g.Coutp / 128);
   auto s = at::cuda::getCurrentCUDAStream();
-  const size_t lds = 3 * 8192;
+  const size_t lds = 2 * 32768;
   auto* px = reinterpret_cast<const fp8e4*>(xc.data_ptr());
   auto* pw = reinterpret_cast<const fp8e4*>(wpk.data_ptr());
   const fp8e4* pz = reinterpret_cast<const fp8e4*>(zero_page_bf16(xc));

@@ -48,7 +48,7 @@ def eval_map(net, items, img, conf_th=0.15, topk=20):
     with torch.no_grad(), amp.autocast(True):
         for i0 in range(0, img.shape[0], 32):
             boxes, clss, scores = pred(img[i0:i0 + 32])
-            for j in range(boxes.shape[0]):
+            for j in range(len(boxes)):
                 i = i0 + j
                 _, gtb, gtl, voc = items[i]
                 name = voc['annotation']['filename']

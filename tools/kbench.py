@@ -85,6 +85,36 @@ def bench_conv(iters):
                 break
 
 
+def bench_fp8(iters):
+    shapes = [
+        ('3x3 128ch @128^2 B8', 8, 128, 128, 128, 3, 1, 1),
+        ('3x3 128ch @256^2 B8', 8, 128, 128, 256, 3, 1, 1),
+        ('3x3 64->128 @256^2 B8', 8, 64, 128, 256, 3, 1, 1),
+        ('1x1 128ch @128^2 B8', 8, 128, 128, 128, 1, 1, 0),
+    ]
+    for name, B, cin, cout, hw, k, stride, pad in shapes:
+        x8 = (torch.randn(B, cin, hw, hw, device='cuda') * 0.5) \
+            .to(torch.float8_e4m3fn).contiguous(memory_format=CL)
+        w = torch.randn(cout, cin, k, k, device='cuda') * 0.05
+        wpk8 = C.pack_weights_fp8(w)
+        ones = torch.ones(cout, device='cuda')
+        zeros = torch.zeros(cout, device='cuda')
+        fl = 2 * B * hw * hw * cin * cout * k * k
+        ms = timeit(lambda: C.conv_fwd_fp8r(x8, wpk8, ones, zeros, None,
+                                            k, k, stride, pad, cout, 1,
+                                            True), iters)
+        print(f'conv(fp8 K128) {name:24s} {ms*1e3:8.1f}us  '
+              f'{fl/ms/1e9:7.1f} TF')
+        # bf16 twin on the same shape for the A/B
+        xb = torch.randn(B, cin, hw, hw, device='cuda',
+                         dtype=torch.bfloat16).contiguous(memory_format=CL)
+        wpk = C.pack_weights(w, False, True)
+        ms = timeit(lambda: C.conv_fwd(xb, wpk, ones, zeros, None, k, k,
+                                       stride, pad, cout, 1), iters)
+        print(f'conv(bf16)     {name:24s} {ms*1e3:8.1f}us  '
+              f'{fl/ms/1e9:7.1f} TF')
+
+
 def bench_stem(iters):
     x = torch.randn(16, 3, 512, 512, device='cuda',
                     dtype=torch.bfloat16).contiguous(memory_format=CL)
@@ -138,6 +168,8 @@ if __name__ == '__main__':
         bench_wgrad(args.iters)
     if args.which in ('conv', 'all'):
         bench_conv(args.iters)
+    if args.which in ('fp8', 'all'):
+        bench_fp8(args.iters)
     if args.which in ('stem', 'stem_wgrad', 'all'):
         bench_stem(args.iters)
     if args.which in ('bn', 'all'):

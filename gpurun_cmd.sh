@@ -1,9 +1,6 @@
 cd /root/repo
-python -m pytest tests -m gpu -x -q > gpurun_out/r2g_pytest.log 2>&1
-echo "pytest rc=$?"; tail -2 gpurun_out/r2g_pytest.log
-timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph --fp8 > gpurun_out/r2g_infer_fp8.json 2>&1
-echo fp8:; tail -1 gpurun_out/r2g_infer_fp8.json
-timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/r2g_infer_b8.json 2>&1
-echo bf16:; tail -1 gpurun_out/r2g_infer_b8.json
-timeout 600 python tools/quality_probe.py holdout --steps 3000 --train-imgs 256 --val-imgs 64 --eval-every 500 > gpurun_out/r2g_holdout.log 2>&1
-echo "holdout rc=$?"; grep -v libdrm gpurun_out/r2g_holdout.log | head -8
+timeout 180 python tools/kbench.py fp8 --iters 50 > gpurun_out/r2h_kbench_fp8.log 2>&1
+grep conv gpurun_out/r2h_kbench_fp8.log
+cd /tmp && export TMPDIR=/tmp
+timeout 300 rocprofv3 --pmc SQ_WAVE_CYCLES,SQ_ACTIVE_INST_ANY,SQ_WAIT_INST_ANY,SQ_INSTS_MFMA,SQ_INSTS_VALU,SQ_LDS_BANK_CONFLICT --output-format csv -d /root/repo/gpurun_out/pmc_fp8 -o pmc -- python /root/repo/tools/kbench.py fp8 --iters 10 > /root/repo/gpurun_out/r2h_pmc.log 2>&1
+echo "pmc rc=$?"; ls /root/repo/gpurun_out/pmc_fp8/ 2>/dev/null | head -3

@@ -1,6 +1,9 @@
 cd /root/repo
-timeout 180 python tools/kbench.py fp8 --iters 50 > gpurun_out/r2h_kbench_fp8.log 2>&1
-grep conv gpurun_out/r2h_kbench_fp8.log
-cd /tmp && export TMPDIR=/tmp
-timeout 300 rocprofv3 --pmc SQ_WAVE_CYCLES,SQ_ACTIVE_INST_ANY,SQ_WAIT_INST_ANY,SQ_INSTS_MFMA,SQ_INSTS_VALU,SQ_LDS_BANK_CONFLICT --output-format csv -d /root/repo/gpurun_out/pmc_fp8 -o pmc -- python /root/repo/tools/kbench.py fp8 --iters 10 > /root/repo/gpurun_out/r2h_pmc.log 2>&1
-echo "pmc rc=$?"; ls /root/repo/gpurun_out/pmc_fp8/ 2>/dev/null | head -3
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph --fp8 > gpurun_out/r2i_infer_fp8.json 2>&1
+echo fp8:; tail -1 gpurun_out/r2i_infer_fp8.json
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/r2i_infer_b8.json 2>&1
+echo bf16:; tail -1 gpurun_out/r2i_infer_b8.json
+timeout 180 python -m pytest tests/test_gpu_e2e.py -x -q > gpurun_out/r2i_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/r2i_pytest.log
+timeout 180 python bench.py --mode infer --batch-size 1 --steps 100 --warmup 20 --graph > gpurun_out/r2i_infer_b1.json 2>&1
+echo b1:; tail -1 gpurun_out/r2i_infer_b1.json

@@ -62,6 +62,11 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
                              int64_t KH, int64_t KW, int64_t stride,
                              int64_t pad, int64_t Cout, int64_t act,
                              int64_t splitk);
+torch::Tensor conv_fwd_k64(torch::Tensor x, torch::Tensor wpk,
+                           torch::Tensor scale, torch::Tensor shift,
+                           c10::optional<torch::Tensor> skip,
+                           int64_t KH, int64_t KW, int64_t stride,
+                           int64_t pad, int64_t Cout, int64_t act);
 torch::Tensor stem_im2col(torch::Tensor x, int64_t KS, int64_t stride,
                           int64_t pad);
 torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
@@ -105,6 +110,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd_fp8r", &rthd::conv_fwd_fp8r);
   m.def("conv_fwd", &rthd::conv_fwd);
   m.def("conv_fwd_small", &rthd::conv_fwd_small);
+  m.def("conv_fwd_k64", &rthd::conv_fwd_k64);
   m.def("stem_im2col", &rthd::stem_im2col);
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);

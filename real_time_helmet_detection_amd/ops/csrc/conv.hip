@@ -74,7 +74,9 @@ torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16) {
   const int Rows = swap ? Cin : Cout;
   const int Kdim = swap ? Cout : Cin;
   const int Rp = (int)cdiv(Rows, 128) * 128;
-  const int Kp = (int)cdiv(Kdim, 32) * 32;
+  // K pads to 64 so the 64-ch-per-step kernel variant shares the same
+  // packed image as the 32-ch ones (pad region is zeros)
+  const int Kp = (int)cdiv(Kdim, 64) * 64;
   auto opt = wc.options().dtype(to_bf16 ? at::kBFloat16 : at::kFloat);
   auto pk = torch::empty({T_, Rp, Kp}, opt);
   const int64_t n = (int64_t)T_ * Rp * Kp;
@@ -463,6 +465,11 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
                              int64_t KH, int64_t KW, int64_t stride,
                              int64_t pad, int64_t Cout, int64_t act,
                              int64_t splitk);
+torch::Tensor conv_fwd_k64(torch::Tensor x, torch::Tensor wpk,
+                           torch::Tensor scale, torch::Tensor shift,
+                           c10::optional<torch::Tensor> skip,
+                           int64_t KH, int64_t KW, int64_t stride,
+                           int64_t pad, int64_t Cout, int64_t act);
 
 // per-device cached 16-B zero page for the glds out-of-range source (a
 // fresh torch::zeros({8}) per conv call was ~85 FillFunctor launches per
@@ -546,7 +553,7 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
   g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
   g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
   g.Cout = Cout;
-  g.Cinp = (int)cdiv(g.Cin, 32) * 32;
+  g.Cinp = (int)cdiv(g.Cin, 64) * 64;  // pack_weights 64-pad
   g.Coutp = (int)cdiv(Cout, 128) * 128;
   g.M = g.B * g.Ho * g.Wo;
   TORCH_CHECK(g.Cin >= 1, "conv_fwd: bad Cin");
@@ -577,11 +584,11 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
                 "Cin=", g.Cin, " runs the f32 path");
 
     // -------- per-shape variant selection (autotune cache) --------
-    // tune when the 128-tile kernel can't fill 256 CUs x 2 waves, or when
-    // Cout <= 64 (the 128-wide N tile computes >= half dead columns, e.g.
-    // the 152->64 stem GEMM)
+    // variants: 0 = this 128x128/K32 kernel, 1 = 64x64 split-K (small
+    // spatial / Cout<=64 fill), 2 = 128x128/K64 double-buffer (fewer
+    // barriers+glds issues per channel). First eligible call measures.
     const int big_blocks = (int)cdiv(g.M, 128) * (g.Coutp / 128);
-    if (big_blocks < 512 || g.Cout <= 64) {
+    {
       ConvChoice ch{0, 1};
       bool have = false;
       const uint64_t key = conv_key(g);
@@ -591,30 +598,42 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
         if (it != g_conv_tune.end()) { ch = it->second; have = true; }
       }
       if (!have) {
-        // candidate split-K factors for the 64x64 kernel: enough z-blocks
-        // to reach ~1024 blocks, chunk >= 2 K-steps
+        // split-K candidates only where fill is the problem
         const int nsteps = (int)(KH * KW) * (g.Cinp / 32);
         const int base64 = (int)cdiv(g.M, 64) * (int)cdiv(g.Cout, 64);
         std::vector<int> cands;
-        for (int sk : {1, 2, 4, 8, 16}) {
-          if (sk > 1 && (nsteps + sk - 1) / sk < 2) break;
-          if ((int64_t)base64 * sk > 16384) break;
-          cands.push_back(sk);
+        if (big_blocks < 512 || g.Cout <= 64) {
+          for (int sk : {1, 2, 4, 8, 16}) {
+            if (sk > 1 && (nsteps + sk - 1) / sk < 2) break;
+            if ((int64_t)base64 * sk > 16384) break;
+            cands.push_back(sk);
+          }
         }
-        if (cands.empty()) {
-          ch = {0, 1};
-        } else if (stream_capturing(s) || getenv("RTHD_NO_AUTOTUNE")) {
-          // heuristic: smallest sk that brings the launch to >= 768
-          // blocks (or the largest allowed)
-          ch = {1, cands.back()};
-          for (int sk : cands) {
-            if (base64 * sk >= 768) { ch = {1, sk}; break; }
+        if (stream_capturing(s) || getenv("RTHD_NO_AUTOTUNE")) {
+          if (big_blocks >= 512 && g.Cout > 64) {
+            ch = {0, 1};
+          } else if (!cands.empty()) {
+            ch = {1, cands.back()};
+            for (int sk : cands) {
+              if (base64 * sk >= 768) { ch = {1, sk}; break; }
+            }
           }
         } else {
-          float best = 1e30f;
+          float best;
+          {  // time the big kernel: temporarily pin choice to big
+            std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+            g_conv_tune[key] = ConvChoice{0, 1};
+          }
+          best = time_usec([&] {
+            (void)conv_fwd(xc, wpk, scale, shift, skip, KH, KW, stride,
+                           pad, Cout, act);
+          }, s.stream());
           ch = {0, 1};
-          // big-kernel candidate measured via a recursive call with the
-          // cache primed (set, measure, restore)
+          float tk = time_usec([&] {
+            (void)conv_fwd_k64(xc, wpk, scale, shift, skip, KH, KW,
+                               stride, pad, Cout, act);
+          }, s.stream());
+          if (tk < best) { best = tk; ch = {2, 1}; }
           for (int sk : cands) {
             float t = time_usec([&] {
               (void)conv_fwd_small(xc, wpk, scale, shift, skip, KH, KW,
@@ -622,22 +641,16 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
             }, s.stream());
             if (t < best) { best = t; ch = {1, sk}; }
           }
-          {  // time the big kernel: temporarily pin choice to big
-            std::lock_guard<std::mutex> lk(g_conv_tune_mu);
-            g_conv_tune[key] = ConvChoice{0, 1};
-          }
-          float tbig = time_usec([&] {
-            (void)conv_fwd(xc, wpk, scale, shift, skip, KH, KW, stride,
-                           pad, Cout, act);
-          }, s.stream());
-          if (tbig < best) { best = tbig; ch = {0, 1}; }
         }
         std::lock_guard<std::mutex> lk(g_conv_tune_mu);
         g_conv_tune[key] = ch;
       }
-      if (ch.small)
+      if (ch.small == 1)
         return conv_fwd_small(xc, wpk, scale, shift, skip, KH, KW, stride,
                               pad, Cout, act, ch.splitk);
+      if (ch.small == 2)
+        return conv_fwd_k64(xc, wpk, scale, shift, skip, KH, KW, stride,
+                            pad, Cout, act);
     }
 
     auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());

@@ -258,7 +258,7 @@ torch::Tensor conv_fwd_small(torch::Tensor x, torch::Tensor wpk,
   g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
   g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
   g.Cout = Cout;
-  g.Cinp = (int)cdiv(g.Cin, 32) * 32;
+  g.Cinp = (int)cdiv(g.Cin, 64) * 64;  // pack_weights 64-pad
   g.Coutp = (int)cdiv(Cout, 128) * 128;  // pack_weights row padding
   g.M = g.B * g.Ho * g.Wo;
   g.nsteps = (int)(KH * KW) * (g.Cinp / 32);

@@ -364,6 +364,31 @@ def test_conv_fwd_small_splitk(cfg):
     assert rel_err(got, want) < 0.03
 
 
+@pytest.mark.parametrize('cfg', [
+    # (B, Cin, Cout, k, pad, H)
+    (4, 128, 128, 3, 1, 32),
+    (2, 64, 128, 3, 1, 24),      # Cin=64 (one zero K-half)
+    (2, 128, 6, 1, 0, 16),       # head
+    (3, 96, 160, 3, 1, 12),      # ragged everything
+])
+def test_conv_fwd_k64(cfg):
+    """64-ch K-step variant vs torch conv on bf16-rounded data."""
+    b, cin, cout, k, pad, h = cfg
+    torch.manual_seed(23)
+    x = torch.randn(b, cin, h, h).to(torch.bfloat16).float()
+    w = (torch.randn(cout, cin, k, k) * 0.05).to(torch.bfloat16).float()
+    skip = torch.randn(b, cout, h, h).to(torch.bfloat16).float()
+    sc = torch.rand(cout) + 0.5
+    sh = torch.randn(cout) * 0.1
+    want = F.relu(F.conv2d(x, w, None, padding=pad)
+                  * sc.view(1, -1, 1, 1) + sh.view(1, -1, 1, 1) + skip)
+    wpk = _C().pack_weights(w.cuda(), False, True)
+    got = _C().conv_fwd_k64(to_gpu(x, torch.bfloat16), wpk, sc.cuda(),
+                            sh.cuda(), to_gpu(skip, torch.bfloat16),
+                            k, k, 1, pad, cout, 1)
+    assert rel_err(got, want) < 0.03
+
+
 def test_conv_autotune_dispatch_matches_big():
     """The autotuned conv_fwd on a small-M shape must agree with the
     explicit variants (whichever the cache picked)."""

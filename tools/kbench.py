@@ -71,6 +71,14 @@ def bench_conv(iters):
                                        stride, pad, cout, 1), iters)
         print(f'conv(auto) {name:26s} {ms*1e3:8.1f}us  {fl/ms/1e9:7.1f} TF')
         # explicit variants for the table
+        try:
+            ms = timeit(lambda: C.conv_fwd_k64(x, wpk, ones, zeros, None,
+                                               k, k, stride, pad, cout, 1),
+                        iters)
+            print(f'conv(k64)  {name:26s} {ms*1e3:8.1f}us  '
+                  f'{fl/ms/1e9:7.1f} TF')
+        except RuntimeError as e:
+            print(f'conv(k64) {name}: {e}')
         for sk in (1, 2, 4, 8, 16):
             if sk > 9 * max(1, cin // 32):
                 continue

@@ -1,9 +1,10 @@
+cd /tmp && export TMPDIR=/tmp
+timeout 300 rocprofv3 --kernel-trace --stats --output-format csv -d /root/repo/gpurun_out/prof_r2final -o prof -- python /root/repo/bench.py --steps 5 --warmup 2 --no-train-graph > /root/repo/gpurun_out/r2l_prof.log 2>&1
+echo "prof rc=$?"
 cd /root/repo
-timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph --fp8 > gpurun_out/r2i_infer_fp8.json 2>&1
-echo fp8:; tail -1 gpurun_out/r2i_infer_fp8.json
-timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/r2i_infer_b8.json 2>&1
-echo bf16:; tail -1 gpurun_out/r2i_infer_b8.json
-timeout 180 python -m pytest tests/test_gpu_e2e.py -x -q > gpurun_out/r2i_pytest.log 2>&1
-echo "pytest rc=$?"; tail -1 gpurun_out/r2i_pytest.log
-timeout 180 python bench.py --mode infer --batch-size 1 --steps 100 --warmup 20 --graph > gpurun_out/r2i_infer_b1.json 2>&1
-echo b1:; tail -1 gpurun_out/r2i_infer_b1.json
+timeout 300 python bench.py --steps 20 --warmup 5 --num-stack 2 --increase-ch 128 > gpurun_out/r2l_bench_big.json 2>&1
+echo big:; tail -1 gpurun_out/r2l_bench_big.json
+timeout 120 python tools/kbench.py wgrad --iters 40 > gpurun_out/r2l_kbench_wgrad.log 2>&1
+grep wgrad gpurun_out/r2l_kbench_wgrad.log
+timeout 240 python bench.py --steps 30 --warmup 5 --dtype fp32 --no-train-graph > gpurun_out/r2l_bench_fp32.json 2>&1
+echo fp32:; tail -1 gpurun_out/r2l_bench_fp32.json

@@ -212,7 +212,14 @@ static void run_reduce_partials(const float* p1, const float* p2,
                                 float* o1, float* o2, int chunks, int C,
                                 const torch::TensorOptions& opt,
                                 hipStream_t s) {
-  const int ks = std::min(8, chunks);
+  if (chunks <= 8) {
+    // stage 1 would be a pure copy (each y-block owns exactly one row):
+    // reduce the partials directly in fixed order
+    hipLaunchKernelGGL(reduce_final_kernel, dim3(cdiv(C, 256)), dim3(256),
+        0, s, p1, p2, o1, o2, chunks, C);
+    return;
+  }
+  const int ks = 8;
   auto st = torch::empty({(int64_t)2 * ks * C}, opt);
   float* s1 = st.data_ptr<float>();
   float* s2 = p2 ? s1 + (int64_t)ks * C : nullptr;

@@ -141,9 +141,10 @@ class GraphedTrainStep:
                 self.opt.zero_grad(set_to_none=True)
                 return None
             self.graphs[key] = entry
-            # the capture already consumed this batch (warmup+capture)
-            self._log_replay(entry)
-            return entry.hm_logits
+            # stream capture RECORDS without executing: the static loss
+            # tensors hold garbage until the first replay, and the batch
+            # has not had its (post-warmup) update — fall through to the
+            # replay below (statics already hold this batch)
         for dst, src in zip(entry.statics, (image, hm, off, wh, mask)):
             dst.copy_(src, non_blocking=True)
         entry.graph.replay()

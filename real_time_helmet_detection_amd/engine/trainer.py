@@ -149,6 +149,13 @@ def _want_train_graph(args, device):
     sequence)."""
     device = torch.device(device) if not isinstance(device, torch.device) \
         else device
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    # multi-rank multiscale would capture at DIFFERENT iterations per rank
+    # (per-rank random batch sizes -> divergent shape keys): one rank
+    # recording its all-reduce while another executes eagerly deadlocks
+    # the collective sequence — run eagerly instead
+    if world > 1 and getattr(args, 'multiscale_flag', False):
+        return False
     return (getattr(args, 'train_graph', True)
             and device.type == 'cuda'
             and args.train_flag

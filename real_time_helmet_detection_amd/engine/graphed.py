@@ -85,11 +85,12 @@ class GraphedTrainStep:
 
     def _capture(self, batch):
         statics = tuple(t.clone() for t in batch)
-        n_pending = len(self.calc._pending)
+        # warmup steps are real weight updates on this batch but are NOT
+        # logged (the body computes losses directly, bypassing the
+        # LossCalculator history)
         for _ in range(self.WARMUP_STEPS):
             self._eager_body(*statics)
         torch.cuda.synchronize()
-        n_pending_warm = len(self.calc._pending)
         g = torch.cuda.CUDAGraph()
         # thread_local capture mode: the DataLoader pin-memory thread (and
         # the RCCL watchdog) make HIP calls concurrently with the capture;
@@ -98,9 +99,6 @@ class GraphedTrainStep:
         # the moment a loader is attached)
         with torch.cuda.graph(g, capture_error_mode='thread_local'):
             losses, per_stack, hm_logits = self._eager_body(*statics)
-        # warmup entries are real updates (keep); capture-time entries
-        # reference graph-owned memory that every replay overwrites (drop)
-        del self.calc._pending[n_pending_warm:]
         # replays never run python: keep the inference-fold caches honest
         from ..ops import hip
         hip.bump_train_stamp()
@@ -110,7 +108,6 @@ class GraphedTrainStep:
         e.losses = losses
         e.per_stack = per_stack
         e.hm_logits = hm_logits
-        del n_pending
         return e
 
     # ------------------------------------------------------------------

@@ -45,7 +45,9 @@ struct WgradGeo2 {
 
 #define WGRAD_NUM_XCD 8
 
-template <bool ALIGNED>  // Cin,Cout % 8 == 0: branch-free staging loads
+// ALIGNED: Cin,Cout % 8 == 0 (branch-free staging loads); SAME: stride-1
+// same-size conv (compile-time single-mul pixel addressing)
+template <bool ALIGNED, bool SAME>
 __global__ __launch_bounds__(256)
 void wgrad_bf16_kernel(const bf16* __restrict__ x,
                        const bf16* __restrict__ dy,
@@ -67,6 +69,11 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
   const int co0 = blockIdx.y * 64;
   const int dyt = t / g.KW - g.pad;
   const int dxt = t % g.KW - g.pad;
+  // stride-1 same-size convs (every 3x3/1x1 in the model): the input
+  // pixel index is just m + dtoff, collapsing the per-element 3-level
+  // int64 address chain to one mul (the staging phase is VALU-bound:
+  // VALU:MFMA was 16.6:1 in the round-1 PMC capture)
+  const int dtoff = dyt * g.W + dxt;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -118,9 +125,9 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
         /* unconditional load from a clamped address + select-zero: a        \
            branch around the load makes hipcc drain vmcnt(0) per element     \
            (guide trap (c) - measured 2-6x on this kernel) */                \
-        const int64_t off = val                                              \
-            ? (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase          \
-            : 0;                                                             \
+        const int64_t off = !val ? 0                                         \
+            : SAME ? (int64_t)(m + dtoff) * g.Cin + cbase                    \
+            : (((int64_t)bj * g.H + iy) * g.W + ix) * g.Cin + cbase;         \
         v = *reinterpret_cast<const uint4*>(x + off);                        \
         if (!val) v = uint4{0, 0, 0, 0};                                     \
       } else {                                                               \
@@ -198,6 +205,9 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
       if (stage_x) RTHD_WG_LOADX(p0 + 128) else RTHD_WG_LOADY(p0 + 128)
     }
 
+    // raise issue priority for the MFMA phase: co-resident waves still in
+    // their (VALU-heavy) staging phase yield issue slots to the MFMAs
+    asm volatile("s_setprio 1");
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
       const int k8 = (lane >> 4) + ks * 4;
@@ -218,6 +228,7 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               xa[mi], yb[ni], acc[mi][ni], 0, 0, 0);
     }
+    asm volatile("s_setprio 0");
   }
 
 #undef RTHD_WG_LOADX
@@ -328,16 +339,17 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();
   const bool aligned = (g.Cin % 8 == 0) && (g.Cout % 8 == 0);
-  if (aligned)
-    hipLaunchKernelGGL((wgrad_bf16_kernel<true>), grid, dim3(256), 0, s,
-        reinterpret_cast<const bf16*>(xc.data_ptr()),
-        reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        dwp.data_ptr<float>(), g);
-  else
-    hipLaunchKernelGGL((wgrad_bf16_kernel<false>), grid, dim3(256), 0, s,
-        reinterpret_cast<const bf16*>(xc.data_ptr()),
-        reinterpret_cast<const bf16*>(dyc.data_ptr()),
-        dwp.data_ptr<float>(), g);
+  const bool same = stride == 1 && g.Ho == g.H && g.Wo == g.W;
+  auto* px = reinterpret_cast<const bf16*>(xc.data_ptr());
+  auto* pdy = reinterpret_cast<const bf16*>(dyc.data_ptr());
+#define RTHD_WG_LAUNCH(A_, S_)                                           \
+  hipLaunchKernelGGL((wgrad_bf16_kernel<A_, S_>), grid, dim3(256), 0, s, \
+      px, pdy, dwp.data_ptr<float>(), g)
+  if (aligned && same) RTHD_WG_LAUNCH(true, true);
+  else if (aligned)    RTHD_WG_LAUNCH(true, false);
+  else if (same)       RTHD_WG_LAUNCH(false, true);
+  else                 RTHD_WG_LAUNCH(false, false);
+#undef RTHD_WG_LAUNCH
   const int rblocks = (int)std::min<int64_t>(2048, cdiv(N, 4 * 256) + 1);
   hipLaunchKernelGGL(wgrad_reduce_kernel, dim3(rblocks), dim3(256), 0, s,
       dwp.data_ptr<float>(), dw.data_ptr<float>(), nchunks, N);

@@ -1,7 +1,7 @@
 cd /root/repo
-python -m pytest tests -m gpu -x -q > gpurun_out/r2m_pytest.log 2>&1
-echo "pytest rc=$?"; tail -2 gpurun_out/r2m_pytest.log
-timeout 240 python bench.py --steps 30 --warmup 5 > gpurun_out/r2m_bench1.json 2>&1
-echo train:; tail -1 gpurun_out/r2m_bench1.json
-timeout 300 python main.py --train-flag --synthetic --synthetic-size 512 --batch-size 16 --amp --end-epoch 2 --print-interval 8 --num-workers 8 --save-path /tmp/wg > gpurun_out/r2m_cli.log 2>&1
-echo "cli rc=$?"; grep 'Loss' gpurun_out/r2m_cli.log | tail -3
+timeout 120 python tools/kbench.py wgrad --iters 40 > gpurun_out/r2n_wgrad.log 2>&1
+grep wgrad gpurun_out/r2n_wgrad.log
+timeout 240 python bench.py --steps 30 --warmup 5 > gpurun_out/r2n_bench1.json 2>&1
+echo train:; tail -1 gpurun_out/r2n_bench1.json
+python -m pytest tests/test_gpu_kernels.py tests/test_gpu_e2e.py -x -q > gpurun_out/r2n_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/r2n_pytest.log

@@ -297,8 +297,11 @@ class StackedHourglass(nn.Module):
             prediction = self.head_lst[i](feature)
             predictions.append(prediction)
             if i < self.num_stack - 1:
-                x = x + self.merge_feature[i](feature) \
-                    + self.merge_prediction[i](prediction)
+                # x + merge_feature(feature) + merge_prediction(prediction)
+                # (reference hourglass.py:234-235) with both adds fused
+                # into the 1x1 convs' epilogues
+                mf = self.merge_feature[i](feature, skip=x)
+                x = self.merge_prediction[i](prediction, skip=mf)
         return torch.stack(predictions, dim=1)
 
 

@@ -1,10 +1,11 @@
 cd /root/repo
-python -m pytest tests -m gpu -x -q > gpurun_out/r2o_pytest.log 2>&1
-echo "pytest rc=$?"; tail -1 gpurun_out/r2o_pytest.log
-timeout 300 python bench.py --steps 20 --warmup 5 --num-stack 2 --increase-ch 128 > gpurun_out/r2o_big.json 2>&1
-echo big:; tail -1 gpurun_out/r2o_big.json
-timeout 300 python bench.py --steps 20 --warmup 5 --num-stack 2 > gpurun_out/r2o_2stack.json 2>&1
-echo 2stack:; tail -1 gpurun_out/r2o_2stack.json
-cd /tmp && export TMPDIR=/tmp
-timeout 300 rocprofv3 --kernel-trace --stats --output-format csv -d /root/repo/gpurun_out/prof_b1 -o p -- python /root/repo/bench.py --mode infer --batch-size 1 --steps 50 --warmup 10 > /root/repo/gpurun_out/r2o_b1prof.log 2>&1
-echo "b1 prof rc=$?"; tail -2 /root/repo/gpurun_out/r2o_b1prof.log | head -1
+python -m pytest tests -m gpu -x -q > gpurun_out/r2p_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/r2p_pytest.log
+timeout 180 python bench.py --mode infer --batch-size 1 --steps 100 --warmup 20 --graph > gpurun_out/r2p_b1.json 2>&1
+echo b1:; tail -1 gpurun_out/r2p_b1.json
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/r2p_b8.json 2>&1
+echo b8:; tail -1 gpurun_out/r2p_b8.json
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph --fp8 > gpurun_out/r2p_fp8.json 2>&1
+echo fp8:; tail -1 gpurun_out/r2p_fp8.json
+timeout 180 python bench.py --mode infer --batch-size 1 --steps 100 --warmup 20 > gpurun_out/r2p_b1_nograph.json 2>&1
+echo b1-nograph:; tail -1 gpurun_out/r2p_b1_nograph.json

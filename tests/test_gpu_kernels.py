@@ -248,6 +248,22 @@ def test_compute_stack_losses_gpu_matches_cpu():
 
 # ----------------------------------------------------------------- decode --
 
+def test_nms_overflow_falls_back_to_eager():
+    """>2048 boxes exceed the LDS-resident kernel: the wrapper must fall
+    back to the eager suppression instead of aborting (reference used
+    torchvision.ops.nms with no cap — e.g. --topk 1000, num_stack 3)."""
+    from real_time_helmet_detection_amd.ops import eager, hip
+    torch.manual_seed(30)
+    n = 3000
+    ctr = torch.rand(n, 2) * 400
+    wh2 = torch.rand(n, 2) * 60 + 5
+    boxes = torch.cat([ctr - wh2, ctr + wh2], dim=1)
+    scores = torch.rand(n)
+    want = eager.nms(boxes, scores, 0.5)
+    got = hip.nms(boxes.cuda(), scores.cuda(), 0.5)
+    assert got.cpu().tolist() == want.tolist()
+
+
 def test_decode_matches_eager():
     from real_time_helmet_detection_amd.ops import eager
     torch.manual_seed(6)

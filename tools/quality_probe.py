@@ -40,12 +40,14 @@ def load_split(n, size, seed):
                    *(t.cuda() for t in (hm, off, wh, mask)))
 
 
-def eval_map(net, items, img, conf_th=0.15, topk=20):
+def eval_map(net, items, img, conf_th=0.15, topk=20, fp8=False):
     net.eval()
     pred = Prediction(net, topk=topk, scale_factor=4, conf_th=conf_th,
                       nms='nms', nms_th=0.5).cuda()
     gt, preds = {}, {}
-    with torch.no_grad(), amp.autocast(True):
+    import contextlib
+    fp8_ctx = amp.fp8_autocast(True) if fp8 else contextlib.nullcontext()
+    with torch.no_grad(), amp.autocast(True), fp8_ctx:
         for i0 in range(0, img.shape[0], 32):
             boxes, clss, scores = pred(img[i0:i0 + 32])
             for j in range(len(boxes)):
@@ -121,6 +123,11 @@ def main():
                 row['train_map@0.5'] = round(
                     eval_map(net, tr_items[:args.val_imgs],
                              img[:args.val_imgs]), 4)
+                if args.in_ch % 128 == 0:
+                    # fp8-resident serving quality on the same held-out
+                    # split (VERDICT round-1 item 6: fp8 mAP probe)
+                    row['val_map@0.5_fp8'] = round(
+                        eval_map(net, val_items, val_img, fp8=True), 4)
             curve.append(row)
             print(json.dumps(row), flush=True)
     print(json.dumps({'mode': args.mode, 'final': curve[-1] if curve

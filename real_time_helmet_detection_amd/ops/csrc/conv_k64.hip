@@ -36,7 +36,7 @@ DEV_INLINE int k64_off(int row, int chunk) {        // 128-B rows, 16-B chunks
   return row * 128 + ((chunk ^ (row & 7)) << 4);
 }
 
-template <bool HAS_SKIP>
+template <bool HAS_SKIP, bool STATS = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_bf16_k64_kernel(const bf16* __restrict__ x,
                               const bf16* __restrict__ wpk,
@@ -45,7 +45,9 @@ void conv_fwd_bf16_k64_kernel(const bf16* __restrict__ x,
                               const bf16* __restrict__ skip,
                               const bf16* __restrict__ zpage,
                               bf16* __restrict__ y,
-                              ConvGeoK64 g, int act) {
+                              ConvGeoK64 g, int act,
+                              float* __restrict__ sp1 = nullptr,
+                              float* __restrict__ sp2 = nullptr) {
   const int mblk = blockIdx.x;
   const int nblk = blockIdx.y;
   const int tid = threadIdx.x;
@@ -171,6 +173,7 @@ void conv_fwd_bf16_k64_kernel(const bf16* __restrict__ x,
     esh[ni] = c < g.Cout ? shift[c] : 0.f;
   }
   const int row_in_frag = (lane >> 4) * 4;
+  float s1[4] = {}, s2[4] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -186,9 +189,68 @@ void conv_fwd_bf16_k64_kernel(const bf16* __restrict__ x,
         if (HAS_SKIP) v += ldf(&skip[(int64_t)m * g.Cout + c]);
         v = apply_act(v, act);
         stf(&y[(int64_t)m * g.Cout + c], v);
+        if (STATS) {
+          s1[ni] += v;
+          s2[ni] += v * v;
+        }
       }
     }
   }
+  if (STATS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      float a = s1[ni], b = s2[ni];
+      a += __shfl_xor(a, 16, 64);
+      a += __shfl_xor(a, 32, 64);
+      b += __shfl_xor(b, 16, 64);
+      b += __shfl_xor(b, 32, 64);
+      const int c = col0 + ni * 16;
+      if ((lane >> 4) == 0 && c < g.Cout) {
+        const int64_t chunk = (int64_t)mblk * 2 + wr;
+        sp1[chunk * g.Cout + c] = a;
+        sp2[chunk * g.Cout + c] = b;
+      }
+    }
+  }
+}
+
+// k64 with the fused-stats epilogue (no skip — training BN runs the conv
+// linear; skip/act fuse later in bn_act_fwd). p1/p2: [2*Mblks, Cout] f32.
+torch::Tensor conv_fwd_k64_stats(torch::Tensor x, torch::Tensor wpk,
+                                 torch::Tensor scale, torch::Tensor shift,
+                                 int64_t KH, int64_t KW, int64_t stride,
+                                 int64_t pad, int64_t Cout, int64_t act,
+                                 torch::Tensor p1, torch::Tensor p2) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  if (xc.scalar_type() != at::kBFloat16) xc = xc.to(at::kBFloat16);
+  ConvGeoK64 g;
+  g.B = xc.size(0);
+  g.Cin = xc.size(1);
+  g.H = xc.size(2);
+  g.W = xc.size(3);
+  g.KH = KH; g.KW = KW; g.stride = stride; g.pad = pad;
+  g.Ho = (g.H + 2 * g.pad - (int)KH) / (int)stride + 1;
+  g.Wo = (g.W + 2 * g.pad - (int)KW) / (int)stride + 1;
+  g.Cout = Cout;
+  g.Cinp = (int)cdiv(g.Cin, 64) * 64;
+  g.Coutp = (int)cdiv(Cout, 128) * 128;
+  g.M = g.B * g.Ho * g.Wo;
+  auto y = torch::empty({g.B, (int64_t)g.Cout, g.Ho, g.Wo},
+                        xc.options().memory_format(
+                            at::MemoryFormat::ChannelsLast));
+  auto sc = scale.to(at::kFloat).contiguous();
+  auto sh = shift.to(at::kFloat).contiguous();
+  dim3 grid(cdiv(g.M, 128), g.Coutp / 128);
+  auto s = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((conv_fwd_bf16_k64_kernel<false, true>), grid,
+      dim3(256), 2 * 32768, s,
+      reinterpret_cast<const bf16*>(xc.data_ptr()),
+      reinterpret_cast<const bf16*>(wpk.data_ptr()),
+      sc.data_ptr<float>(), sh.data_ptr<float>(), nullptr,
+      zero_page_bf16(xc), reinterpret_cast<bf16*>(y.data_ptr()), g,
+      (int)act, p1.data_ptr<float>(), p2.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return y;
 }
 
 torch::Tensor conv_fwd_k64(torch::Tensor x, torch::Tensor wpk,

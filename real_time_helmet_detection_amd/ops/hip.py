@@ -148,9 +148,28 @@ class _ConvBNActFn(torch.autograd.Function):
         mean = rstd = y_lin = None
         if use_bn and training:
             # conv output INCLUDES the (redundant-under-BN) bias so running
-            # stats match the eager/reference semantics (stem has bias+BN)
-            y_lin = run_conv(ones, bias_f, ACT_CODE['Linear'])
-            mean, rstd = C.bn_stats(y_lin, rmean, rvar, momentum, eps)
+            # stats match the eager/reference semantics (stem has bias+BN).
+            # On the bf16 MFMA variants the column sum/sumsq partials come
+            # fused from the conv epilogue (no standalone colsum pass).
+            if bf16:
+                if stem_col:
+                    outs = C.conv_fwd_stats(xc, wpk, ones, bias_f, 1, 1,
+                                            1, 0, cout, ACT_CODE['Linear'])
+                else:
+                    outs = C.conv_fwd_stats(xc, wpk, ones, bias_f, kh, kw,
+                                            stride, pad, cout,
+                                            ACT_CODE['Linear'])
+                y_lin = outs[0]
+                if len(outs) == 3:
+                    mean, rstd = C.bn_stats_from_parts(
+                        outs[1], outs[2], rmean, rvar, momentum, eps,
+                        y_lin.numel() // cout)
+                else:
+                    mean, rstd = C.bn_stats(y_lin, rmean, rvar, momentum,
+                                            eps)
+            else:
+                y_lin = run_conv(ones, bias_f, ACT_CODE['Linear'])
+                mean, rstd = C.bn_stats(y_lin, rmean, rvar, momentum, eps)
             y = C.bn_act_fwd(y_lin, mean, rstd, gamma, beta, act_code, skc)
         elif use_bn:
             rstd_run = torch.rsqrt(rvar.float() + eps)

@@ -52,15 +52,25 @@ def _run_pair(make_module, x_shape, seed, train=True, x_grad=True):
 TOL = 2e-4  # fp32 MFMA vs CPU fma-order differences only
 
 
-def _assert_grads(cpu, gpu, xc, xg, yc, yg):
-    assert rel_err(yg, yc) < TOL, 'forward'
-    assert rel_err(xg.grad, xc.grad) < TOL, 'dx'
+def _param_grads_close(cpu, gpu):
     for (name, pc), (_, pg) in zip(cpu.named_parameters(),
                                    gpu.named_parameters()):
         if pc.grad is None:
             assert pg.grad is None or pg.grad.abs().max() == 0
             continue
+        if pc.grad.abs().max() < 1e-4:
+            # analytically-zero grads (conv bias under train-mode BN):
+            # CPU eager leaves ~1e-5 summation residue, the HIP path
+            # returns exact zeros — both must be ~0
+            assert pg.grad.abs().max().item() < 1e-4, f'grad {name}'
+            continue
         assert rel_err(pg.grad, pc.grad) < TOL, f'grad {name}'
+
+
+def _assert_grads(cpu, gpu, xc, xg, yc, yg):
+    assert rel_err(yg, yc) < TOL, 'forward'
+    assert rel_err(xg.grad, xc.grad) < TOL, 'dx'
+    _param_grads_close(cpu, gpu)
 
 
 def test_gradcheck_conv_bn_relu():
@@ -101,11 +111,7 @@ def test_gradcheck_prelayer():
         lambda: PreLayer(in_ch=3, mid_ch=32, out_ch=32),
         (2, 3, 64, 64), seed=45, x_grad=False)
     assert rel_err(yg, yc) < TOL
-    for (name, pc), (_, pg) in zip(cpu.named_parameters(),
-                                   gpu.named_parameters()):
-        if pc.grad is None:
-            continue
-        assert rel_err(pg.grad, pc.grad) < TOL, f'grad {name}'
+    _param_grads_close(cpu, gpu)
 
 
 def test_gradcheck_bn_running_stats_match():

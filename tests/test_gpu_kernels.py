@@ -405,6 +405,38 @@ def test_conv_fwd_k64(cfg):
     assert rel_err(got, want) < 0.03
 
 
+@pytest.mark.parametrize('hw', [16, 48])
+def test_conv_fwd_stats_matches_bn_stats(hw):
+    """Fused-epilogue BN stats == standalone bn_stats on the conv output
+    (and the conv output itself matches conv_fwd)."""
+    torch.manual_seed(24)
+    x = torch.randn(3, 32, hw, hw).to(torch.bfloat16).float()
+    w = (torch.randn(32, 32, 3, 3) * 0.1).to(torch.bfloat16).float()
+    bias = torch.randn(32)
+    wpk = _C().pack_weights(w.cuda(), False, True)
+    ones = torch.ones(32, device='cuda')
+    y_ref = _C().conv_fwd(to_gpu(x, torch.bfloat16), wpk, ones,
+                          bias.cuda(), None, 3, 3, 1, 1, 32, 0)
+    outs = _C().conv_fwd_stats(to_gpu(x, torch.bfloat16), wpk, ones,
+                               bias.cuda(), 3, 3, 1, 1, 32, 0)
+    assert rel_err(outs[0], y_ref) < 1e-6
+    want_mean, want_rstd = _C().bn_stats(y_ref, None, None, 0.1, 1e-5)
+    if len(outs) == 3:
+        rm = torch.zeros(32, device='cuda')
+        rv = torch.ones(32, device='cuda')
+        mean, rstd = _C().bn_stats_from_parts(outs[1], outs[2], rm, rv,
+                                              0.1, 1e-5,
+                                              y_ref.numel() // 32)
+        assert rel_err(mean, want_mean) < 1e-4
+        assert rel_err(rstd, want_rstd) < 1e-4
+        # running stats updated like bn_stats does
+        rm2 = torch.zeros(32, device='cuda')
+        rv2 = torch.ones(32, device='cuda')
+        _C().bn_stats(y_ref, rm2, rv2, 0.1, 1e-5)
+        assert rel_err(rm, rm2) < 1e-4
+        assert rel_err(rv, rv2) < 1e-4
+
+
 def test_conv_autotune_dispatch_matches_big():
     """The autotuned conv_fwd on a small-M shape must agree with the
     explicit variants (whichever the cache picked)."""

@@ -1,5 +1,7 @@
 cd /root/repo
-python -m pytest tests/test_gpu_gradcheck.py -x -q > gpurun_out/r2r_gradcheck.log 2>&1
-echo "gradcheck rc=$?"; tail -3 gpurun_out/r2r_gradcheck.log | head -2
-python -m pytest tests -m gpu -x -q > gpurun_out/r2r_pytest.log 2>&1
-echo "full rc=$?"; tail -1 gpurun_out/r2r_pytest.log
+python -m pytest tests -m gpu -x -q > gpurun_out/r2s_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/r2s_pytest.log
+timeout 240 python bench.py --steps 30 --warmup 5 > gpurun_out/r2s_bench1.json 2>&1
+echo train:; tail -1 gpurun_out/r2s_bench1.json
+timeout 240 python bench.py --steps 20 --warmup 5 --num-stack 2 --increase-ch 128 > gpurun_out/r2s_big.json 2>&1
+echo big:; tail -1 gpurun_out/r2s_big.json

@@ -219,7 +219,11 @@ static void run_reduce_partials(const float* p1, const float* p2,
         0, s, p1, p2, o1, o2, chunks, C);
     return;
   }
-  const int ks = 8;
+  // scale the stage-1 split with the row count: a fixed ks=8 left the
+  // conv-epilogue stats reduce (4096 partial rows @128^2) on 16 blocks
+  int ks = chunks / 16;
+  if (ks < 8) ks = 8;
+  if (ks > 256) ks = 256;
   auto st = torch::empty({(int64_t)2 * ks * C}, opt);
   float* s1 = st.data_ptr<float>();
   float* s2 = p2 ? s1 + (int64_t)ks * C : nullptr;

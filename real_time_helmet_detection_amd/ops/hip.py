@@ -151,7 +151,8 @@ class _ConvBNActFn(torch.autograd.Function):
             # stats match the eager/reference semantics (stem has bias+BN).
             # On the bf16 MFMA variants the column sum/sumsq partials come
             # fused from the conv epilogue (no standalone colsum pass).
-            if bf16:
+            import os as _os
+            if bf16 and _os.environ.get('RTHD_NO_FUSED_STATS') != '1':
                 if stem_col:
                     outs = C.conv_fwd_stats(xc, wpk, ones, bias_f, 1, 1,
                                             1, 0, cout, ACT_CODE['Linear'])

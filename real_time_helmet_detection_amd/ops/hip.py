@@ -149,10 +149,13 @@ class _ConvBNActFn(torch.autograd.Function):
         if use_bn and training:
             # conv output INCLUDES the (redundant-under-BN) bias so running
             # stats match the eager/reference semantics (stem has bias+BN).
-            # On the bf16 MFMA variants the column sum/sumsq partials come
-            # fused from the conv epilogue (no standalone colsum pass).
+            # Fused-epilogue BN stats exist (conv_fwd_stats) but measured
+            # ~6% SLOWER end to end than the standalone colsum pass (the
+            # extra epilogue stores + shuffles extend the conv's
+            # store-tail more than the saved 0.4 ms read pass; same-box
+            # A/B 940 vs 997 img/s) — opt-in via RTHD_FUSED_STATS=1.
             import os as _os
-            if bf16 and _os.environ.get('RTHD_NO_FUSED_STATS') != '1':
+            if bf16 and _os.environ.get('RTHD_FUSED_STATS') == '1':
                 if stem_col:
                     outs = C.conv_fwd_stats(xc, wpk, ones, bias_f, 1, 1,
                                             1, 0, cout, ACT_CODE['Linear'])

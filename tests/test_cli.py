@@ -67,3 +67,26 @@ def test_export_cli(tmp_path):
     with torch.no_grad():
         out = m(torch.randn(1, 3, 64, 64))
     assert len(out) == 3
+
+
+def test_main_train_torchrun_2rank(tmp_path):
+    """The exact multi-rank CLI path the 8-GPU run uses: torchrun launches
+    main.py --train-flag, one worker per rank (gloo on CPU), bucketed
+    all-reduce, rank-0 checkpoint."""
+    env = dict(os.environ)
+    env['PYTHONPATH'] = REPO
+    env.setdefault('MASTER_ADDR', '127.0.0.1')
+    r = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29641', 'main.py'] + TRAIN_FLAGS +
+        ['--save-path', str(tmp_path) + '/'],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=900)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    # rank-0 checkpoint with the reference dict format
+    ckpt_path = tmp_path / 'check_point_1.pth'
+    assert ckpt_path.exists(), r.stdout[-1500:]
+    ckpt = torch.load(ckpt_path, map_location='cpu', weights_only=False)
+    for k in ('epoch', 'state_dict', 'optimizer', 'scheduler', 'scaler',
+              'loss_log'):
+        assert k in ckpt, k

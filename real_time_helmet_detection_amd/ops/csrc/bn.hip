@@ -191,32 +191,62 @@ __global__ void reduce_partials_kernel(const float* __restrict__ p1,
   }
 }
 
+// Optionally fuses the BN finalize (mean/rstd + running-stat update) so
+// bn_stats is one kernel shorter (fin_mean != nullptr selects it).
 __global__ void reduce_final_kernel(const float* __restrict__ s1,
                                     const float* __restrict__ s2,
                                     float* __restrict__ o1,
-                                    float* __restrict__ o2, int K, int C) {
+                                    float* __restrict__ o2, int K, int C,
+                                    float* __restrict__ fin_mean = nullptr,
+                                    float* __restrict__ fin_rstd = nullptr,
+                                    float* __restrict__ run_mean = nullptr,
+                                    float* __restrict__ run_var = nullptr,
+                                    float Mf = 1.f, float momentum = 0.1f,
+                                    float eps = 1e-5f) {
   const int c = blockIdx.x * 256 + threadIdx.x;
   if (c >= C) return;
   float a = 0.f;
   for (int k = 0; k < K; ++k) a += s1[(int64_t)k * C + c];
   o1[c] = a;
+  float b = 0.f;
   if (o2) {
-    float b = 0.f;
     for (int k = 0; k < K; ++k) b += s2[(int64_t)k * C + c];
     o2[c] = b;
+  }
+  if (fin_mean) {
+    const float mu = a / Mf;
+    float var = b / Mf - mu * mu;
+    var = fmaxf(var, 0.f);
+    fin_mean[c] = mu;
+    fin_rstd[c] = rsqrtf(var + eps);
+    if (run_mean) {
+      run_mean[c] = (1.f - momentum) * run_mean[c] + momentum * mu;
+      const float ub = Mf > 1.f ? var * Mf / (Mf - 1.f) : var;
+      run_var[c] = (1.f - momentum) * run_var[c] + momentum * ub;
+    }
   }
 }
 
 // staged two-kernel deterministic reduction of [chunks][C] partials
+struct BnFinalize {
+  float* mean = nullptr;
+  float* rstd = nullptr;
+  float* run_mean = nullptr;
+  float* run_var = nullptr;
+  float Mf = 1.f, momentum = 0.1f, eps = 1e-5f;
+};
+
 static void run_reduce_partials(const float* p1, const float* p2,
                                 float* o1, float* o2, int chunks, int C,
                                 const torch::TensorOptions& opt,
-                                hipStream_t s) {
+                                hipStream_t s,
+                                const BnFinalize& fin = BnFinalize{}) {
   if (chunks <= 8) {
     // stage 1 would be a pure copy (each y-block owns exactly one row):
     // reduce the partials directly in fixed order
     hipLaunchKernelGGL(reduce_final_kernel, dim3(cdiv(C, 256)), dim3(256),
-        0, s, p1, p2, o1, o2, chunks, C);
+        0, s, p1, p2, o1, o2, chunks, C, fin.mean, fin.rstd, fin.run_mean,
+        fin.run_var, fin.Mf, fin.momentum, fin.eps);
     return;
   }
   // scale the stage-1 split with the row count: a fixed ks=8 left the
@@ -230,31 +260,10 @@ static void run_reduce_partials(const float* p1, const float* p2,
   hipLaunchKernelGGL(reduce_partials_kernel, dim3(cdiv(C, 64), ks),
       dim3(256), 0, s, p1, p2, s1, s2, chunks, C);
   hipLaunchKernelGGL(reduce_final_kernel, dim3(cdiv(C, 256)), dim3(256),
-      0, s, s1, s2, o1, o2, ks, C);
+      0, s, s1, s2, o1, o2, ks, C, fin.mean, fin.rstd, fin.run_mean,
+      fin.run_var, fin.Mf, fin.momentum, fin.eps);
 }
 
-// finalize mean/rstd (+ running-stat update, torch semantics)
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
-                                   float* __restrict__ mean,
-                                   float* __restrict__ rstd,
-                                   float* __restrict__ running_mean,
-                                   float* __restrict__ running_var,
-                                   int C, float Mf, float momentum,
-                                   float eps) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float mu = sum[c] / Mf;
-  float var = sumsq[c] / Mf - mu * mu;
-  var = fmaxf(var, 0.f);
-  mean[c] = mu;
-  rstd[c] = rsqrtf(var + eps);
-  if (running_mean) {
-    const float unbiased = Mf > 1.f ? var * Mf / (Mf - 1.f) : var;
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
-  }
-}
 
 // ---------------------------- bn apply forward ------------------------------
 
@@ -546,7 +555,8 @@ __global__ void bn_act_bwd_apply8_kernel(
 
 static void run_colsum(const torch::Tensor& xc, torch::Tensor& sum,
                        torch::Tensor* sumsq, int64_t M, int C,
-                       hipStream_t s) {
+                       hipStream_t s,
+                       const BnFinalize& fin = BnFinalize{}) {
   const int chunks = pick_chunks(M, C);
   auto p1 = torch::empty({chunks, C}, sum.options());
   torch::Tensor p2;
@@ -578,7 +588,7 @@ static void run_colsum(const torch::Tensor& xc, torch::Tensor& sum,
   }
   run_reduce_partials(p1.data_ptr<float>(), p2p, sum.data_ptr<float>(),
                       sumsq ? sumsq->data_ptr<float>() : nullptr, chunks,
-                      C, sum.options(), s);
+                      C, sum.options(), s, fin);
 }
 
 std::vector<torch::Tensor> bn_stats(torch::Tensor x,
@@ -594,18 +604,18 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   auto mean = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto rstd = torch::empty({C}, xc.options().dtype(at::kFloat));
   auto s = at::cuda::getCurrentCUDAStream();
-  run_colsum(xc, sum, &sumsq, M, C, s);
-  float* rm = nullptr;
-  float* rv = nullptr;
+  BnFinalize fin;
+  fin.mean = mean.data_ptr<float>();
+  fin.rstd = rstd.data_ptr<float>();
   if (running_mean.has_value()) {
     TORCH_CHECK(running_mean->scalar_type() == at::kFloat);
-    rm = running_mean->data_ptr<float>();
-    rv = running_var->data_ptr<float>();
+    fin.run_mean = running_mean->data_ptr<float>();
+    fin.run_var = running_var->data_ptr<float>();
   }
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0, s,
-      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-      mean.data_ptr<float>(), rstd.data_ptr<float>(), rm, rv, C, (float)M,
-      (float)momentum, (float)eps);
+  fin.Mf = (float)M;
+  fin.momentum = (float)momentum;
+  fin.eps = (float)eps;
+  run_colsum(xc, sum, &sumsq, M, C, s, fin);  // finalize fused in
   HIP_CHECK_LAST();
   return {mean, rstd};
 }
@@ -628,20 +638,20 @@ std::vector<torch::Tensor> bn_stats_from_parts(
   auto mean = torch::empty({C}, opt);
   auto rstd = torch::empty({C}, opt);
   auto s = at::cuda::getCurrentCUDAStream();
-  run_reduce_partials(p1.data_ptr<float>(), p2.data_ptr<float>(),
-                      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-                      chunks, C, opt, s);
-  float* rm = nullptr;
-  float* rv = nullptr;
+  BnFinalize fin;
+  fin.mean = mean.data_ptr<float>();
+  fin.rstd = rstd.data_ptr<float>();
   if (running_mean.has_value()) {
     TORCH_CHECK(running_mean->scalar_type() == at::kFloat);
-    rm = running_mean->data_ptr<float>();
-    rv = running_var->data_ptr<float>();
+    fin.run_mean = running_mean->data_ptr<float>();
+    fin.run_var = running_var->data_ptr<float>();
   }
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
-      s, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-      mean.data_ptr<float>(), rstd.data_ptr<float>(), rm, rv, C, (float)M,
-      (float)momentum, (float)eps);
+  fin.Mf = (float)M;
+  fin.momentum = (float)momentum;
+  fin.eps = (float)eps;
+  run_reduce_partials(p1.data_ptr<float>(), p2.data_ptr<float>(),
+                      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                      chunks, C, opt, s, fin);
   HIP_CHECK_LAST();
   return {mean, rstd};
 }

@@ -250,7 +250,8 @@ def main():
             'higher_is_better': True,
             'scaling': 'weak',
             'vs_baseline': None,
-            'dtype': args.dtype if use_cuda else 'fp32',
+            'dtype': ('fp8' if (args.fp8 and args.mode == 'infer')
+                      else args.dtype) if use_cuda else 'fp32',
             'data': 'synthetic',
             'config': {
                 'model': 'hourglass-%d-ch%d%s' % (

@@ -1,6 +1,9 @@
-cd /tmp && export TMPDIR=/tmp
-timeout 300 rocprofv3 --kernel-trace --stats --output-format csv -d /root/repo/gpurun_out/prof_b1v2 -o p -- python /root/repo/bench.py --mode infer --batch-size 1 --steps 60 --warmup 10 > /root/repo/gpurun_out/r2v_b1prof.log 2>&1
-echo "b1 prof rc=$?"
 cd /root/repo
-timeout 900 python tools/quality_probe.py holdout --steps 2000 --train-imgs 192 --val-imgs 48 --eval-every 400 --in-ch 128 --size 512 --batch-size 8 --lr 1e-3 > gpurun_out/r2v_holdout512.log 2>&1
-echo "holdout512 rc=$?"; grep -v libdrm gpurun_out/r2v_holdout512.log | head -6
+python -m pytest tests -m gpu -x -q > gpurun_out/r2w_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/r2w_pytest.log
+timeout 180 python bench.py --mode infer --batch-size 1 --steps 100 --warmup 20 --graph > gpurun_out/r2w_b1.json 2>&1
+echo b1:; tail -1 gpurun_out/r2w_b1.json
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph > gpurun_out/r2w_b8.json 2>&1
+echo b8:; tail -1 gpurun_out/r2w_b8.json
+timeout 240 python bench.py --steps 30 --warmup 5 > gpurun_out/r2w_train.json 2>&1
+echo train:; tail -1 gpurun_out/r2w_train.json

@@ -71,6 +71,20 @@ class Prediction(torch.nn.Module):
         scores = scores.reshape(b, s * self.topk)
 
         box_lst, cls_lst, score_lst = [], [], []
+        if self.nms == 'nms':
+            # batched kernel with the confidence filter folded in: one
+            # launch + one host sync for the whole batch (the per-image
+            # loop paid a device sync per image)
+            idx, counts = ops.nms_batched(boxes, scores, self.nms_th,
+                                          self.conf_th)
+            counts_l = counts.cpu().tolist()
+            for i in range(b):
+                sel = idx[i, :counts_l[i]].long()
+                box_lst.append(boxes[i][sel])
+                cls_lst.append(clss[i][sel])
+                score_lst.append(scores[i][sel])
+            return box_lst, cls_lst, score_lst
+
         for i in range(b):
             keep = scores[i] >= self.conf_th
             bi, ci, si = boxes[i][keep], clss[i][keep], scores[i][keep]
@@ -257,6 +271,16 @@ class GraphedPredictor(torch.nn.Module):
         boxes, clss, scores = (t.clone() for t in self.static_out)
         p = self.p
         box_lst, cls_lst, score_lst = [], [], []
+        if p.nms == 'nms':
+            idx, counts = ops.nms_batched(boxes, scores, p.nms_th,
+                                          p.conf_th)
+            counts_l = counts.cpu().tolist()
+            for i in range(boxes.shape[0]):
+                sel = idx[i, :counts_l[i]].long()
+                box_lst.append(boxes[i][sel])
+                cls_lst.append(clss[i][sel])
+                score_lst.append(scores[i][sel])
+            return box_lst, cls_lst, score_lst
         for i in range(boxes.shape[0]):
             keep = scores[i] >= p.conf_th
             bi, ci, si = boxes[i][keep], clss[i][keep], scores[i][keep]

@@ -53,6 +53,13 @@ def nms(boxes, scores, iou_threshold):
     return eager.nms(boxes, scores, iou_threshold)
 
 
+def nms_batched(boxes, scores, iou_threshold, conf_th):
+    if _hip(boxes):
+        from . import hip
+        return hip.nms_batched(boxes, scores, iou_threshold, conf_th)
+    return eager.nms_batched(boxes, scores, iou_threshold, conf_th)
+
+
 def soft_nms(boxes, scores, iou_threshold=0.3, sigma=0.5, score_th=0.001):
     # O(N^2) sequential rescoring on <=few hundred boxes: host-side everywhere.
     return eager.soft_nms(boxes, scores, iou_threshold, sigma, score_th)

@@ -42,6 +42,10 @@ std::vector<torch::Tensor> decode_fwd(torch::Tensor hm, torch::Tensor off,
                                       bool normalized);
 torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
                       double iou_threshold);
+std::vector<torch::Tensor> nms_batched(torch::Tensor boxes,
+                                       torch::Tensor scores,
+                                       double iou_threshold,
+                                       double conf_th);
 
 torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16);
 torch::Tensor pack_weights_fp8(torch::Tensor w);
@@ -114,6 +118,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("centernet_loss_fused_bwd", &rthd::centernet_loss_fused_bwd);
   m.def("decode_fwd", &rthd::decode_fwd);
   m.def("nms_fwd", &rthd::nms_fwd);
+  m.def("nms_batched", &rthd::nms_batched);
   m.def("pack_weights", &rthd::pack_weights);
   m.def("pack_weights_fp8", &rthd::pack_weights_fp8);
   m.def("conv_fwd_fp8r", &rthd::conv_fwd_fp8r);
@@ -155,6 +160,11 @@ static torch::Tensor avgpool2x2_op(torch::Tensor x) {
 static torch::Tensor maxpool_same_op(torch::Tensor x, int64_t k) {
   return maxpool_same_fwd(std::move(x), k, /*need_arg=*/false)[0];
 }
+static std::tuple<torch::Tensor, torch::Tensor> nms_batched_op(
+    torch::Tensor boxes, torch::Tensor scores, double iou, double conf) {
+  auto v = nms_batched(std::move(boxes), std::move(scores), iou, conf);
+  return {v[0], v[1]};
+}
 static std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> decode_op(
     torch::Tensor hm, torch::Tensor off, torch::Tensor wh,
     int64_t scale_factor, int64_t topk, int64_t pool_size, bool normalized) {
@@ -178,6 +188,8 @@ TORCH_LIBRARY(rthd, m) {
   m.def("maxpool_same(Tensor x, int k) -> Tensor");
   m.def("upsample2x_add(Tensor x, Tensor? skip) -> Tensor");
   m.def("nms(Tensor boxes, Tensor scores, float iou) -> Tensor");
+  m.def("nms_batched(Tensor boxes, Tensor scores, float iou, float conf) "
+        "-> (Tensor, Tensor)");
   m.def("decode(Tensor hm, Tensor off, Tensor wh, int scale_factor, "
         "int topk, int pool_size, bool normalized) "
         "-> (Tensor, Tensor, Tensor)");
@@ -193,6 +205,7 @@ TORCH_LIBRARY_IMPL(rthd, CUDA, m) {
   m.impl("maxpool_same", rthd::maxpool_same_op);
   m.impl("upsample2x_add", rthd::upsample2x_add_fwd);
   m.impl("nms", rthd::nms_fwd);
+  m.impl("nms_batched", rthd::nms_batched_op);
   m.impl("decode", rthd::decode_op);
   m.impl("stem_im2col", rthd::stem_im2col);
 }

@@ -25,12 +25,22 @@ namespace rthd {
 
 constexpr int NMS_CAP = 2048;
 
-__global__ void nms_kernel(const float* __restrict__ boxes,  // (N,4)
-                           const float* __restrict__ scores, // (N,)
-                           int* __restrict__ out_idx,        // (N,)
-                           int* __restrict__ out_count,
-                           unsigned long long* __restrict__ mask_g,
-                           int N, int P, int W, float thr) {
+// grid.x = image index (batched form; the single-image entry uses B=1).
+// conf: candidates with score < conf are neither suppressors nor output
+// (equivalent to filtering before NMS — they sort to the tail).
+__global__ void nms_kernel(const float* __restrict__ boxes,  // (B,N,4)
+                           const float* __restrict__ scores, // (B,N)
+                           int* __restrict__ out_idx,        // (B,N)
+                           int* __restrict__ out_count,      // (B,)
+                           unsigned long long* __restrict__ mask_gb,
+                           int N, int P, int W, float thr, float conf) {
+  const int bimg = blockIdx.x;
+  boxes += (int64_t)bimg * N * 4;
+  scores += (int64_t)bimg * N;
+  out_idx += (int64_t)bimg * N;
+  out_count += bimg;
+  unsigned long long* mask_g =
+      mask_gb ? mask_gb + (int64_t)bimg * N * W : nullptr;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* sx1 = reinterpret_cast<float*>(smem);
   float* sy1 = sx1 + P;
@@ -119,6 +129,7 @@ __global__ void nms_kernel(const float* __restrict__ boxes,  // (N,4)
   if (threadIdx.x < 64) {
     const int w = threadIdx.x;
     for (int i = 0; i < N; ++i) {
+      if (ss[i] < conf) break;  // sorted desc: tail is all below conf
       const bool alive = ((rem[i >> 6] >> (i & 63)) & 1ull) == 0ull;
       if (alive && w < W) rem[w] |= mask[(int64_t)i * W + w];
     }
@@ -127,11 +138,64 @@ __global__ void nms_kernel(const float* __restrict__ boxes,  // (N,4)
 
   if (threadIdx.x == 0) {
     int k = 0;
-    for (int i = 0; i < N; ++i)
+    for (int i = 0; i < N && ss[i] >= conf; ++i)
       if (((rem[i >> 6] >> (i & 63)) & 1ull) == 0ull)
         out_idx[k++] = sidx[i];
     *out_count = k;
   }
+}
+
+// shared launch logic for the single-image and batched entries
+static void launch_nms(const float* pb, const float* psc, int* pidx,
+                       int* pcnt, int B, int N, float thr, float conf,
+                       const torch::TensorOptions& opt) {
+  int P = 64;
+  while (P < N) P <<= 1;
+  const int W = (int)cdiv(N, 64);
+  const bool lds_mask = N <= 1024;  // [N][W] mask fits beside the arrays
+  size_t lds = (size_t)P * 24 + (size_t)W * 8;
+  if (lds_mask) lds += (size_t)N * W * 8;
+  torch::Tensor mask_ws;
+  unsigned long long* mg = nullptr;
+  if (!lds_mask) {
+    mask_ws = torch::empty({(int64_t)B * N * W}, opt.dtype(at::kLong));
+    mg = reinterpret_cast<unsigned long long*>(mask_ws.data_ptr());
+  }
+  if (lds > 65536) {
+    static std::once_flag once;
+    std::call_once(once, [] {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&nms_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    });
+  }
+  auto s = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(nms_kernel, dim3(B), dim3(256), lds, s,
+      pb, psc, pidx, pcnt, mg, N, P, W, thr, conf);
+  HIP_CHECK_LAST();
+}
+
+// batched: one kernel, one later host sync for ALL images (the per-image
+// loop in Prediction paid a sync per image). conf folds the confidence
+// filter into the kernel (uniform N per image).
+std::vector<torch::Tensor> nms_batched(torch::Tensor boxes,
+                                       torch::Tensor scores,
+                                       double iou_threshold,
+                                       double conf_th) {
+  auto b = boxes.to(at::kFloat).contiguous();
+  auto sc = scores.to(at::kFloat).contiguous();
+  TORCH_CHECK(b.dim() == 3 && sc.dim() == 2 && b.size(0) == sc.size(0) &&
+              b.size(1) == sc.size(1), "nms_batched: (B,N,4)/(B,N)");
+  const int B = b.size(0), N = b.size(1);
+  TORCH_CHECK(N <= NMS_CAP, "nms_batched: N > cap");
+  auto out_idx = torch::zeros({B, std::max(N, 1)},
+                              b.options().dtype(at::kInt));
+  auto out_count = torch::zeros({B}, b.options().dtype(at::kInt));
+  if (N == 0) return {out_idx, out_count};
+  launch_nms(b.data_ptr<float>(), sc.data_ptr<float>(),
+             out_idx.data_ptr<int>(), out_count.data_ptr<int>(), B, N,
+             (float)iou_threshold, (float)conf_th, b.options());
+  return {out_idx, out_count};
 }
 
 torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
@@ -146,32 +210,9 @@ torch::Tensor nms_fwd(torch::Tensor boxes, torch::Tensor scores,
                               b.options().dtype(at::kInt));
   auto out_count = torch::zeros({1}, b.options().dtype(at::kInt));
   if (N == 0) return torch::empty({0}, b.options().dtype(at::kLong));
-  int P = 64;
-  while (P < N) P <<= 1;
-  const int W = (int)cdiv(N, 64);
-  const bool lds_mask = N <= 1024;  // [N][W] mask fits beside the arrays
-  size_t lds = (size_t)P * 24 + (size_t)W * 8;
-  if (lds_mask) lds += (size_t)N * W * 8;
-  torch::Tensor mask_ws;
-  unsigned long long* mg = nullptr;
-  if (!lds_mask) {
-    mask_ws = torch::empty({(int64_t)N * W},
-                           b.options().dtype(at::kLong));
-    mg = reinterpret_cast<unsigned long long*>(mask_ws.data_ptr());
-  }
-  if (lds > 65536) {
-    static std::once_flag once;
-    std::call_once(once, [] {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&nms_kernel),
-          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-    });
-  }
-  auto s = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(nms_kernel, dim3(1), dim3(256), lds, s,
-      b.data_ptr<float>(), sc.data_ptr<float>(), out_idx.data_ptr<int>(),
-      out_count.data_ptr<int>(), mg, N, P, W, (float)iou_threshold);
-  HIP_CHECK_LAST();
+  launch_nms(b.data_ptr<float>(), sc.data_ptr<float>(),
+             out_idx.data_ptr<int>(), out_count.data_ptr<int>(), 1, N,
+             (float)iou_threshold, -3.4e38f, b.options());
   const int k = out_count.item<int>();  // host sync: result used on host
   return out_idx.narrow(0, 0, k).to(at::kLong);
 }

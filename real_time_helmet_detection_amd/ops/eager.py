@@ -164,6 +164,25 @@ def nms(boxes, scores, iou_threshold):
     return order[keep_mask]
 
 
+def nms_batched(boxes, scores, iou_threshold, conf_th):
+    """Eager twin of the batched NMS kernel: per image, filter by
+    conf_th then greedy NMS; returns (idx (B,N) int32 of KEPT original
+    indices, counts (B) int32)."""
+    B, N = scores.shape
+    idx = torch.zeros(B, N, dtype=torch.int32, device=boxes.device)
+    counts = torch.zeros(B, dtype=torch.int32, device=boxes.device)
+    for i in range(B):
+        keep_conf = scores[i] >= conf_th
+        sel = keep_conf.nonzero(as_tuple=False).squeeze(1)
+        if sel.numel() == 0:
+            continue
+        kept = nms(boxes[i][sel], scores[i][sel], iou_threshold)
+        orig = sel[kept]
+        counts[i] = orig.numel()
+        idx[i, :orig.numel()] = orig.to(torch.int32)
+    return idx, counts
+
+
 def soft_nms(boxes, scores, iou_threshold=0.3, sigma=0.5, score_th=0.001):
     """Gaussian soft-NMS (reference evaluate.py:184-243 capability).
 

@@ -668,6 +668,19 @@ def batched_decode(heatmap, offset, wh, scale_factor, topk, pool_size,
 _NMS_CAP = 2048  # LDS-resident kernel limit (nms.hip)
 
 
+def nms_batched(boxes, scores, iou_threshold, conf_th):
+    """Batched class-agnostic NMS with the confidence filter folded in:
+    one kernel + ONE host sync for the whole batch (the per-image loop
+    paid a sync per image). boxes (B,N,4), scores (B,N) ->
+    (idx (B,N) int32, counts (B) int32)."""
+    if boxes.shape[1] > _NMS_CAP:
+        from .eager import nms_batched as eager_nb
+        return eager_nb(boxes, scores, iou_threshold, conf_th)
+    idx, counts = _C().nms_batched(boxes, scores, float(iou_threshold),
+                                   float(conf_th))
+    return idx, counts
+
+
 def nms(boxes, scores, iou_threshold):
     if boxes.shape[0] > _NMS_CAP:
         # configs like --topk 1000 with num_stack>=3 and conf_th=0 exceed

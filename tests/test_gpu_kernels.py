@@ -264,6 +264,27 @@ def test_nms_overflow_falls_back_to_eager():
     assert got.cpu().tolist() == want.tolist()
 
 
+def test_nms_batched_matches_eager():
+    """Batched NMS kernel (conf filter folded in) vs the eager per-image
+    filter-then-NMS loop — exact index/count match."""
+    from real_time_helmet_detection_amd.ops import eager
+    torch.manual_seed(33)
+    B, N = 5, 300
+    ctr = torch.rand(B, N, 2) * 400
+    wh2 = torch.rand(B, N, 2) * 60 + 5
+    boxes = torch.cat([ctr - wh2, ctr + wh2], dim=2)
+    scores = torch.rand(B, N)
+    for conf in (0.0, 0.35):
+        want_idx, want_cnt = eager.nms_batched(boxes, scores, 0.5, conf)
+        got_idx, got_cnt = _C().nms_batched(boxes.cuda(), scores.cuda(),
+                                            0.5, conf)
+        assert got_cnt.cpu().tolist() == want_cnt.tolist()
+        for i in range(B):
+            k = int(want_cnt[i])
+            assert got_idx[i, :k].cpu().tolist() == \
+                want_idx[i, :k].tolist(), (conf, i)
+
+
 def test_decode_matches_eager():
     from real_time_helmet_detection_amd.ops import eager
     torch.manual_seed(6)

@@ -438,11 +438,15 @@ def conv_bn_act(x, conv, bn, act, act_module=None, training=False,
     if act_code is None:
         raise NotImplementedError(f'HIP conv epilogue: activation {act!r}')
     if not training and not torch.is_grad_enabled():
-        # fp8 K=128 MFMA path: only when the channel count fills whole
-        # 128-k blocks — a Cin=64 layer would burn half its MFMA work on
-        # zero padding (measured 409 vs 492 TF against bf16), so the
-        # stem-adjacent 64-ch conv stays bf16
-        if _amp.fp8_enabled() and not is_stem and cin % 128 == 0:
+        # fp8 K=128 MFMA path: only when (a) the channel count fills
+        # whole 128-k blocks (a Cin=64 layer would burn half its MFMA on
+        # zero padding — measured 409 vs 492 TF) and (b) the GEMM is big
+        # enough to fill the chip with 128x128 tiles — small-spatial
+        # layers run the bf16 path, whose autotune has 64-tile/split-K
+        # variants (the fp8 kernel does not); the extra dtype boundary
+        # casts there are on tiny tensors
+        if (_amp.fp8_enabled() and not is_stem and cin % 128 == 0
+                and x.shape[0] * x.shape[2] * x.shape[3] >= 16384):
             y = _conv_infer_fp8(x, conv, bn, act_code, skip)
         else:
             y = _conv_infer(x, conv, bn, act_code, skip, kh, kw, stride,

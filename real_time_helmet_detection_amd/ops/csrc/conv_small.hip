@@ -151,12 +151,14 @@ void conv_fwd_bf16_64_kernel(const bf16* __restrict__ x,
       bfrag[i] = *reinterpret_cast<const bf16x8*>(
           B + lds_off_bf16_s(brow_base + 16 * i, k8));
     }
+    asm volatile("s_setprio 1");
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    asm volatile("s_setprio 0");
 
     if (step + 2 < nsteps + 1) {
       if (step + 2 < nsteps)

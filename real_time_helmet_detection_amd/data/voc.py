@@ -168,7 +168,8 @@ class SyntheticVOC(_CollateMixin, torch.utils.data.Dataset):
         return self.size
 
     def __getitem__(self, index):
-        rng = np.random.RandomState(self.seed * 100003 + index)
+        rng = np.random.RandomState((self.seed * 100003 + index)
+                                    % (2 ** 32))
         h = w = self.imsize
         img = rng.randint(0, 128, size=(h, w, 3), dtype=np.uint8)
         n = rng.randint(1, 7)

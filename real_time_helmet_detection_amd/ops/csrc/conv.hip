@@ -37,10 +37,15 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // swap=false: pk[t][co][ci] = w[co][ci][t/KW][t%KW]
 // swap=true (dgrad): roles swapped + taps rotated:
 //   pk[t][ci][co] = w[co][ci][KH-1-t/KW][KW-1-t%KW]  (rows indexed by ci)
+// reads w through explicit element strides so channels_last master
+// weights pack WITHOUT a contiguous() relayout first (that copy ran
+// twice per conv per training step — ~0.5 ms/step)
 template <typename T>
 __global__ void pack_weights_kernel(const float* __restrict__ w,
                                     T* __restrict__ pk,
                                     int Cout, int Cin, int KH, int KW,
+                                    int64_t s0, int64_t s1, int64_t s2,
+                                    int64_t s3,
                                     int Rows, int Rp, int Kp, int swap) {
   // Rows/Rp: row count (+pad) of pk (= Cout or Cin); Kp: padded k per tap
   const int T_ = KH * KW;
@@ -60,14 +65,14 @@ __global__ void pack_weights_kernel(const float* __restrict__ w,
       } else {
         ci = row; co = k; ty = KH - 1 - t / KW; tx = KW - 1 - t % KW;
       }
-      v = w[(((int64_t)co * Cin + ci) * KH + ty) * KW + tx];
+      v = w[co * s0 + ci * s1 + ty * s2 + tx * s3];
     }
     stf(&pk[i], v);
   }
 }
 
 torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16) {
-  auto wc = w.to(at::kFloat).contiguous();  // (Cout, Cin, KH, KW)
+  auto wc = w.to(at::kFloat);  // (Cout, Cin, KH, KW), any stride layout
   const int Cout = wc.size(0), Cin = wc.size(1);
   const int KH = wc.size(2), KW = wc.size(3);
   const int T_ = KH * KW;
@@ -81,15 +86,17 @@ torch::Tensor pack_weights(torch::Tensor w, bool swap, bool to_bf16) {
   auto pk = torch::empty({T_, Rp, Kp}, opt);
   const int64_t n = (int64_t)T_ * Rp * Kp;
   auto s = at::cuda::getCurrentCUDAStream();
+  const auto ws = wc.strides();
   if (to_bf16)
     hipLaunchKernelGGL((pack_weights_kernel<bf16>), dim3(ew_grid(n, 256)),
         dim3(256), 0, s, wc.data_ptr<float>(),
         reinterpret_cast<bf16*>(pk.data_ptr()), Cout, Cin, KH, KW,
-        Rows, Rp, Kp, swap ? 1 : 0);
+        ws[0], ws[1], ws[2], ws[3], Rows, Rp, Kp, swap ? 1 : 0);
   else
     hipLaunchKernelGGL((pack_weights_kernel<float>), dim3(ew_grid(n, 256)),
         dim3(256), 0, s, wc.data_ptr<float>(), pk.data_ptr<float>(),
-        Cout, Cin, KH, KW, Rows, Rp, Kp, swap ? 1 : 0);
+        Cout, Cin, KH, KW, ws[0], ws[1], ws[2], ws[3], Rows, Rp, Kp,
+        swap ? 1 : 0);
   HIP_CHECK_LAST();
   return pk;
 }
@@ -815,6 +822,8 @@ std::vector<torch::Tensor> conv_fwd_stats(
 __global__ void pack_weights_fp8_kernel(const float* __restrict__ w,
                                         unsigned char* __restrict__ pk,
                                         int Cout, int Cin, int KH, int KW,
+                                        int64_t s0, int64_t s1, int64_t s2,
+                                        int64_t s3,
                                         int Rp, int Kp) {
   const int T_ = KH * KW;
   const int64_t n = (int64_t)T_ * Rp * Kp;
@@ -826,14 +835,14 @@ __global__ void pack_weights_fp8_kernel(const float* __restrict__ w,
     const int k = i % Kp;
     float v = 0.f;
     if (row < Cout && k < Cin)
-      v = w[(((int64_t)row * Cin + k) * KH + t / KW) * KW + t % KW];
+      v = w[row * s0 + k * s1 + (t / KW) * s2 + (t % KW) * s3];
     const unsigned p = __builtin_amdgcn_cvt_pk_fp8_f32(v, 0.f, 0, false);
     pk[i] = (unsigned char)(p & 0xff);
   }
 }
 
 torch::Tensor pack_weights_fp8(torch::Tensor w) {
-  auto wc = w.to(at::kFloat).contiguous();
+  auto wc = w.to(at::kFloat);
   const int Cout = wc.size(0), Cin = wc.size(1);
   const int KH = wc.size(2), KW = wc.size(3);
   const int T_ = KH * KW;
@@ -844,10 +853,11 @@ torch::Tensor pack_weights_fp8(torch::Tensor w) {
                          wc.options().dtype(at::kFloat8_e4m3fn));
   const int64_t n = (int64_t)T_ * Rp * Kp;
   auto s = at::cuda::getCurrentCUDAStream();
+  const auto ws = wc.strides();
   hipLaunchKernelGGL(pack_weights_fp8_kernel, dim3(ew_grid(n, 256)),
       dim3(256), 0, s, wc.data_ptr<float>(),
       reinterpret_cast<unsigned char*>(pk.data_ptr()), Cout, Cin, KH, KW,
-      Rp, Kp);
+      ws[0], ws[1], ws[2], ws[3], Rp, Kp);
   HIP_CHECK_LAST();
   return pk;
 }

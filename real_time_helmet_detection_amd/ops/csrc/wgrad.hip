@@ -175,7 +175,7 @@ void wgrad_kernel(const T* __restrict__ x, const T* __restrict__ dy,
       for (int ni = 0; ni < 2; ++ni) {
         const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
-        dwc[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx] =
+        dwc[(((int64_t)co * g.KH + ty) * g.KW + tx) * g.Cin + ci] =
             acc[mi][ni][r];
       }
     }
@@ -225,8 +225,10 @@ torch::Tensor wgrad(torch::Tensor x, torch::Tensor dy, int64_t KH,
   const int64_t N = (int64_t)g.Cout * g.Cin * KH * KW;
   auto dwp = torch::empty({(int64_t)nchunks * N},
                           xc.options().dtype(at::kFloat));
+  // channels_last: matches the (CL-converted) conv weight layout
   auto dw = torch::empty({g.Cout, g.Cin, KH, KW},
-                         xc.options().dtype(at::kFloat));
+                         xc.options().dtype(at::kFloat).memory_format(
+                             at::MemoryFormat::ChannelsLast));
 
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();

@@ -250,7 +250,10 @@ void wgrad_bf16_kernel(const bf16* __restrict__ x,
       for (int ni = 0; ni < 2; ++ni) {
         const int co = co0 + wc * 32 + ni * 16 + (lane & 15);
         if (co >= g.Cout) continue;
-        dwc[(((int64_t)co * g.Cin + ci) * g.KH + ty) * g.KW + tx] =
+        // channels_last element order: the reduced dw is returned as a
+        // CL tensor so AccumulateGrad assigns it to the CL conv weight
+        // without a relayout copy
+        dwc[(((int64_t)co * g.KH + ty) * g.KW + tx) * g.Cin + ci] =
             acc[mi][ni][r2];
       }
     }
@@ -334,7 +337,8 @@ torch::Tensor wgrad_bf16_fast(torch::Tensor x, torch::Tensor dy, int64_t KH,
   auto dwp = torch::empty({(int64_t)nchunks * N},
                           xc.options().dtype(at::kFloat));
   auto dw = torch::empty({g.Cout, g.Cin, KH, KW},
-                         xc.options().dtype(at::kFloat));
+                         xc.options().dtype(at::kFloat).memory_format(
+                             at::MemoryFormat::ChannelsLast));
 
   dim3 grid(ci_tiles, co_tiles, taps * nchunks);
   auto s = at::cuda::getCurrentCUDAStream();

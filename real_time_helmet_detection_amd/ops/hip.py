@@ -271,7 +271,8 @@ class _ConvBNActFn(torch.autograd.Function):
                         memory_format=torch.channels_last) if cout_p                         else dpre
                     dw = C.wgrad_bf16_fast(xs, ds, kh, kw, stride, pad)
                     if cin_p or cout_p:
-                        dw = dw[:dpre.shape[1], :xc.shape[1]].contiguous()
+                        dw = dw[:dpre.shape[1], :xc.shape[1]].contiguous(
+                            memory_format=torch.channels_last)
                 else:
                     dw = C.wgrad(xc, dpre, kh, kw, stride, pad)
             cur.wait_stream(side)

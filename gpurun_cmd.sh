@@ -1,5 +1,9 @@
 cd /root/repo
-timeout 1450 python main.py --train-flag --synthetic --synthetic-size 7581 --batch-size 16 --amp --end-epoch 100 --num-workers 12 --print-interval 400 --save-path /tmp/full > gpurun_out/final_fulltrain.log 2>&1
-echo "train rc=$?"; tail -2 gpurun_out/final_fulltrain.log
-timeout 300 python main.py --synthetic --synthetic-size 1000 --random-seed 4242 --model-load /tmp/full/check_point_100.pth --save-path /tmp/full --conf-th 0.15 > gpurun_out/final_fulleval.log 2>&1
-echo "eval rc=$?"; grep -iE 'map|ap ' gpurun_out/final_fulleval.log | tail -5
+python -m pytest tests -m gpu -x -q > gpurun_out/fin2_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/fin2_pytest.log
+python __graft_entry__.py smoke > gpurun_out/fin2_smoke.log 2>&1
+echo "smoke rc=$?"; tail -1 gpurun_out/fin2_smoke.log
+for i in 1 2; do
+  timeout 240 python bench.py --steps 30 --warmup 5 2>/dev/null | tail -1 | python3 -c "import json,sys; print('train:', json.loads(sys.stdin.read())['value'])"
+done
+timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph 2>/dev/null | tail -1 | python3 -c "import json,sys; print('b8:', json.loads(sys.stdin.read())['value'])"

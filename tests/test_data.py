@@ -155,3 +155,55 @@ def test_train_augmentor_end_to_end():
             assert (b[:, 2] >= b[:, 0]).all() and (b[:, 3] >= b[:, 1]).all()
             assert (b >= 0).all()
             assert (b <= out_imgs[0].shape[0]).all()
+
+
+def test_train_augmentor_worker_rng_diverges(monkeypatch):
+    """Forked DataLoader workers must NOT share one augmentation stream
+    (round-1 ADVICE: the parent-process RandomState was inherited
+    identically by every worker)."""
+    from real_time_helmet_detection_amd.data.augment import TrainAugmentor
+    import torch.utils.data as tud
+
+    class _FakeInfo:
+        def __init__(self, seed):
+            self.seed = seed
+
+    draws = []
+    for wseed in (1000, 2000):
+        monkeypatch.setattr(tud, 'get_worker_info', lambda s=wseed:
+                            _FakeInfo(s))
+        aug = TrainAugmentor(seed=7)
+        draws.append([aug.rng.uniform() for _ in range(4)])
+    assert draws[0] != draws[1]
+
+    # and the same worker seed reproduces the same stream
+    monkeypatch.setattr(tud, 'get_worker_info', lambda: _FakeInfo(1000))
+    aug = TrainAugmentor(seed=7)
+    assert [aug.rng.uniform() for _ in range(4)] == draws[0]
+
+
+def test_parse_voc_xml_shapes():
+    """Rewritten XML parser: single object stays a LIST under
+    annotation, repeated tags become lists, empty leaves are dropped."""
+    import xml.etree.ElementTree as ET
+    from real_time_helmet_detection_amd.data.voc import (parse_voc_xml,
+                                                         boxes_from_voc_dict)
+    one = ET.fromstring(
+        '<annotation><filename>a.jpg</filename><size><width>10</width>'
+        '<height>20</height></size><object><name>hat</name><bndbox>'
+        '<xmin>1</xmin><ymin>2</ymin><xmax>5</xmax><ymax>6</ymax>'
+        '</bndbox></object><empty></empty></annotation>')
+    d = parse_voc_xml(one)
+    assert isinstance(d['annotation']['object'], list)
+    assert d['annotation']['size']['width'] == '10'
+    assert 'empty' not in d['annotation']
+    boxes, labels = boxes_from_voc_dict(d)
+    assert boxes == [[1, 2, 5, 6]] and labels == [0]
+
+    two = ET.fromstring(
+        '<annotation><object><name>hat</name><bndbox><xmin>1</xmin>'
+        '<ymin>1</ymin><xmax>2</xmax><ymax>2</ymax></bndbox></object>'
+        '<object><name>person</name><bndbox><xmin>3</xmin><ymin>3</ymin>'
+        '<xmax>4</xmax><ymax>4</ymax></bndbox></object></annotation>')
+    boxes, labels = boxes_from_voc_dict(parse_voc_xml(two))
+    assert labels == [0, 1] and len(boxes) == 2

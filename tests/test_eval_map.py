@@ -111,3 +111,22 @@ def test_write_detection_txt(tmp_path):
     txt = (tmp_path / 'img1.txt').read_text().strip()
     assert txt == '0 0.900000 10 20 30 40'
     assert (tmp_path / 'img2.txt').read_text() == ''
+
+
+def test_nms_batched_eager_matches_per_image():
+    """Eager batched-NMS twin == filter-then-NMS per image (the CPU
+    Prediction path now routes through it)."""
+    import torch
+    from real_time_helmet_detection_amd.ops import eager
+    torch.manual_seed(9)
+    B, N = 3, 60
+    ctr = torch.rand(B, N, 2) * 200
+    wh2 = torch.rand(B, N, 2) * 40 + 4
+    boxes = torch.cat([ctr - wh2, ctr + wh2], dim=2)
+    scores = torch.rand(B, N)
+    idx, counts = eager.nms_batched(boxes, scores, 0.5, 0.3)
+    for i in range(B):
+        keep_conf = scores[i] >= 0.3
+        sel = keep_conf.nonzero(as_tuple=False).squeeze(1)
+        want = sel[eager.nms(boxes[i][sel], scores[i][sel], 0.5)]
+        assert idx[i, :counts[i]].tolist() == want.tolist()

@@ -1,9 +1,12 @@
 cd /root/repo
-python -m pytest tests -m gpu -x -q > gpurun_out/fin2_pytest.log 2>&1
-echo "pytest rc=$?"; tail -1 gpurun_out/fin2_pytest.log
-python __graft_entry__.py smoke > gpurun_out/fin2_smoke.log 2>&1
-echo "smoke rc=$?"; tail -1 gpurun_out/fin2_smoke.log
-for i in 1 2; do
-  timeout 240 python bench.py --steps 30 --warmup 5 2>/dev/null | tail -1 | python3 -c "import json,sys; print('train:', json.loads(sys.stdin.read())['value'])"
-done
-timeout 180 python bench.py --mode infer --batch-size 8 --steps 50 --warmup 10 --graph 2>/dev/null | tail -1 | python3 -c "import json,sys; print('b8:', json.loads(sys.stdin.read())['value'])"
+python -m pytest tests -m gpu -x -q > gpurun_out/fin3_pytest.log 2>&1
+echo "pytest rc=$?"; tail -1 gpurun_out/fin3_pytest.log
+timeout 700 python tools/bench_all.py --out gpurun_out/fin3_bench_all.json > /dev/null 2>&1
+echo "bench_all rc=$?"
+python3 -c "
+import json
+r = json.load(open('gpurun_out/fin3_bench_all.json'))
+for k, v in r['configs'].items():
+    res = v.get('result')
+    print(k, '->', (res['value'], res['metric']) if res else v.get('ok', v.get('rc')))
+"

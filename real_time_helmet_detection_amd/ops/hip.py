@@ -21,6 +21,8 @@ MFMA path (guide §3: v_mfma_f32_16x16x4_f32 — identical numerics to an fmaf
 chain).
 """
 
+import os
+
 import torch
 
 from . import _backend
@@ -154,8 +156,7 @@ class _ConvBNActFn(torch.autograd.Function):
             # extra epilogue stores + shuffles extend the conv's
             # store-tail more than the saved 0.4 ms read pass; same-box
             # A/B 940 vs 997 img/s) — opt-in via RTHD_FUSED_STATS=1.
-            import os as _os
-            if bf16 and _os.environ.get('RTHD_FUSED_STATS') == '1':
+            if bf16 and os.environ.get('RTHD_FUSED_STATS') == '1':
                 if stem_col:
                     outs = C.conv_fwd_stats(xc, wpk, ones, bias_f, 1, 1,
                                             1, 0, cout, ACT_CODE['Linear'])

@@ -264,6 +264,36 @@ def test_nms_overflow_falls_back_to_eager():
     assert got.cpu().tolist() == want.tolist()
 
 
+def test_decode_empty_heatmap():
+    """No positive scores anywhere: decode must emit topk zero-score rows
+    (idx 0) without tripping the peak/threshold machinery."""
+    hm = torch.zeros(2, 2, 32, 32, device='cuda')
+    off = torch.rand(2, 2, 32, 32, device='cuda')
+    wh = torch.rand(2, 2, 32, 32, device='cuda')
+    boxes, clss, scores = _C().decode_fwd(hm, off, wh, 4, 10, 3, False)
+    assert scores.abs().max().item() == 0.0
+    assert boxes.shape == (2, 10, 4) and torch.isfinite(boxes).all()
+
+
+def test_nms_batched_global_mask_path():
+    """N > 1024 exceeds the LDS mask budget — the global-scratch branch
+    must match the eager result exactly."""
+    from real_time_helmet_detection_amd.ops import eager
+    torch.manual_seed(35)
+    B, N = 2, 1500
+    ctr = torch.rand(B, N, 2) * 600
+    wh2 = torch.rand(B, N, 2) * 50 + 5
+    boxes = torch.cat([ctr - wh2, ctr + wh2], dim=2)
+    scores = torch.rand(B, N)
+    want_idx, want_cnt = eager.nms_batched(boxes, scores, 0.5, 0.2)
+    got_idx, got_cnt = _C().nms_batched(boxes.cuda(), scores.cuda(),
+                                        0.5, 0.2)
+    assert got_cnt.cpu().tolist() == want_cnt.tolist()
+    for i in range(B):
+        k = int(want_cnt[i])
+        assert got_idx[i, :k].cpu().tolist() == want_idx[i, :k].tolist()
+
+
 def test_nms_batched_matches_eager():
     """Batched NMS kernel (conf filter folded in) vs the eager per-image
     filter-then-NMS loop — exact index/count match."""

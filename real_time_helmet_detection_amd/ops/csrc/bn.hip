@@ -249,11 +249,13 @@ static void run_reduce_partials(const float* p1, const float* p2,
         fin.run_var, fin.Mf, fin.momentum, fin.eps);
     return;
   }
-  // scale the stage-1 split with the row count: a fixed ks=8 left the
-  // conv-epilogue stats reduce (4096 partial rows @128^2) on 16 blocks
-  int ks = chunks / 16;
-  if (ks < 8) ks = 8;
-  if (ks > 256) ks = 256;
+  // ks sets BOTH stage-1 parallelism and the single-block stage-2 loop
+  // length: ks=8 is right for the common <=1024-chunk reduces (scaling
+  // it with chunks made stage 2 latency-bound — measured 5 -> 16 us);
+  // only the opt-in conv-epilogue stats path (up to 4096 rows) needs a
+  // wider stage 1
+  int ks = 8;
+  if (chunks > 1024) ks = std::min(256, chunks / 16);
   auto st = torch::empty({(int64_t)2 * ks * C}, opt);
   float* s1 = st.data_ptr<float>();
   float* s2 = p2 ? s1 + (int64_t)ks * C : nullptr;

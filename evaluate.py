@@ -39,8 +39,13 @@ if __name__ == '__main__':
     cls_lst = cls_ten[0].tolist()
     score_lst = score_ten[0].tolist()
 
-    # clamp to the resized image bounds
-    box_lst = [[max(0, min(v, imsize)) for v in box] for box in box_lst]
+    # clamp to the resized image bounds; order the corners so degenerate
+    # predictions (x2 < x1 after clamping, possible from an undertrained
+    # model) still draw instead of crashing PIL
+    def _sane(box):
+        x1, y1, x2, y2 = (max(0, min(v, imsize)) for v in box)
+        return [min(x1, x2), min(y1, y2), max(x1, x2), max(y1, y2)]
+    box_lst = [_sane(box) for box in box_lst]
 
     img_pil = ten2pil(img_ten[0], args.pretrained)
     for i, (box, cls, score) in enumerate(zip(box_lst, cls_lst, score_lst)):

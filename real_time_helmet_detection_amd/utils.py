@@ -98,10 +98,14 @@ _CLS_COLORS = [(255, 64, 64), (64, 160, 255), (64, 255, 96), (255, 224, 64),
 
 
 def draw_box(image, box, cls=0, width=2):
-    """Draw one xyxy box on a PIL image in the class color."""
+    """Draw one xyxy box on a PIL image in the class color. Corners are
+    ordered first — degenerate predictions (x2 < x1) must not crash the
+    renderer."""
     draw = ImageDraw.Draw(image)
     color = _CLS_COLORS[int(cls) % len(_CLS_COLORS)]
-    draw.rectangle(list(map(float, box)), outline=color, width=width)
+    x1, y1, x2, y2 = map(float, box)
+    draw.rectangle([min(x1, x2), min(y1, y2), max(x1, x2), max(y1, y2)],
+                   outline=color, width=width)
     return image
 
 

@@ -88,3 +88,13 @@ def test_grad_scaler_api_compat():
     assert w.grad is not None
     sd = sc.state_dict()
     sc.load_state_dict(sd)
+
+
+def test_draw_box_degenerate():
+    """Inverted-corner boxes (undertrained predictions) must render, not
+    raise (PIL requires x1 >= x0)."""
+    from PIL import Image
+    from real_time_helmet_detection_amd.utils import draw_box
+    img = Image.new('RGB', (64, 64))
+    draw_box(img, [40.0, 40.0, 10.0, 12.0], cls=1)
+    draw_box(img, [5, 5, 5, 5], cls=0)

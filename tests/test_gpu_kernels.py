@@ -507,6 +507,23 @@ def test_conv_autotune_dispatch_matches_big():
     assert rel_err(got2, want) < 0.03
 
 
+def test_conv_fwd_fp8_multi_kblock():
+    """Cin=256 (increase-ch configs): two 128-k blocks per tap — the
+    kc>1 staging walk of the fp8 kernel."""
+    torch.manual_seed(27)
+    x8 = (torch.randn(2, 256, 12, 12) * 0.5).to(torch.float8_e4m3fn)
+    w = torch.randn(128, 256, 3, 3) * 0.05
+    sw = w.abs().amax(dim=(1, 2, 3)).clamp(min=1e-8) / 240.0
+    w8 = (w / sw.view(-1, 1, 1, 1)).to(torch.float8_e4m3fn).float() \
+        * sw.view(-1, 1, 1, 1)
+    want = F.conv2d(x8.float(), w8, None, padding=1)
+    wpk = _C().pack_weights_fp8((w / sw.view(-1, 1, 1, 1)).cuda())
+    got = _C().conv_fwd_fp8r(
+        x8.cuda().contiguous(memory_format=CL), wpk, sw.cuda(),
+        torch.zeros(128, device='cuda'), None, 3, 3, 1, 1, 128, 0, False)
+    assert rel_err(got.float(), want) < 0.05
+
+
 @pytest.mark.parametrize('out_fp8', [True, False])
 def test_conv_fwd_fp8_resident(out_fp8):
     """fp8-resident conv (e4m3 in, e4m3/bf16 out, fused epilogue + fp8
